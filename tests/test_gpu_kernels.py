@@ -139,3 +139,19 @@ def test_scale_parameter(ext):
         out, _ = local_attention(q, k, v, softmax_scale=scale)
         ref, _ = flash_res_lse(q.cpu(), k.cpu(), v.cpu(), softmax_scale=scale)
         torch.testing.assert_close(out.cpu(), ref, rtol=2.5e-2, atol=2.5e-2)
+
+
+def test_prefill_kernel_512(ext):
+    _check_decode(1, 4, 4, 512, tq=512, causal=True)
+
+
+def test_prefill_kernel_odd_rows(ext):
+    _check_decode(1, 2, 2, 300, tq=300, causal=True)  # Tq not multiple of 256
+
+
+def test_prefill_kernel_noncausal(ext):
+    _check_decode(1, 2, 2, 384, tq=256, causal=False)
+
+
+def test_prefill_kernel_gqa_1024(ext):
+    _check_decode(1, 8, 2, 1024, tq=1024, causal=True, tol=3e-2)
