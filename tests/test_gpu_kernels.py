@@ -155,3 +155,46 @@ def test_prefill_kernel_noncausal(ext):
 
 def test_prefill_kernel_gqa_1024(ext):
     _check_decode(1, 8, 2, 1024, tq=1024, causal=True, tol=3e-2)
+
+
+def test_probe_tr8_semantics(ext):
+    """ds_read_b64_tr_b8 with lane-linear 8-B addresses: lane l receives
+    column (l&15) of the 8x16 row-major BYTE block of its 16-lane group:
+    out[l][j] = ((l&15) + j*16 + (l>>4)*128) & 0xff on an identity LDS."""
+    got = ext.probe_tr8().cpu().to(torch.int32)
+    lanes = torch.arange(64)
+    expect = ((lanes % 16).unsqueeze(1) + torch.arange(8).unsqueeze(0) * 16
+              + (lanes // 16).unsqueeze(1) * 128) % 256
+    assert got.tolist() == expect.tolist(), f"tr8 map differs:\n{got}"
+
+
+def _check_fp8_decode(b, hq, hkv, t, tq=1, causal=False, seed=0, tol=0.15):
+    from tree_attention_torch_amd.ops.flash import local_attention
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+
+    torch.manual_seed(seed)
+    q = torch.randn(b, hq, tq, 128, device="cuda").bfloat16()
+    k8 = torch.randn(b, hkv, t, 128, device="cuda").to(torch.float8_e4m3fn)
+    v8 = torch.randn(b, hkv, t, 128, device="cuda").to(torch.float8_e4m3fn)
+    q_off = t - tq
+    out, lse = local_attention(q, k8, v8, is_causal=causal, q_offset=q_off)
+    # oracle on the DEQUANTIZED k/v (isolates kernel error from quantization)
+    ref_out, ref_lse = flash_res_lse(
+        q.cpu(), k8.cpu().float(), v8.cpu().float(), is_causal=causal, q_offset=q_off
+    )
+    assert torch.isfinite(out).all()
+    torch.testing.assert_close(lse.cpu(), ref_lse, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(out.cpu(), ref_out, rtol=tol, atol=tol)
+
+
+def test_fp8_decode_small(ext):
+    _check_fp8_decode(1, 2, 2, 256)
+
+
+def test_fp8_decode_gqa_32k(ext):
+    _check_fp8_decode(1, 32, 4, 32768)  # BASELINE config 5 head layout
+
+
+def test_fp8_decode_odd(ext):
+    _check_fp8_decode(1, 4, 4, 1000)
+    _check_fp8_decode(1, 4, 4, 100)

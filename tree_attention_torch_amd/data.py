@@ -45,7 +45,11 @@ def make_data(
     """
     b, h, t, d = shape
     hkv = kv_heads if kv_heads is not None else h
-    td = _DTYPES[dtype] if isinstance(dtype, str) else dtype
+    if isinstance(dtype, str) and dtype == "fp8":
+        # fp8 KV cache (OCP e4m3), bf16 queries — BASELINE config 5
+        q_td, td = torch.bfloat16, torch.float8_e4m3fn
+    else:
+        q_td = td = _DTYPES[dtype] if isinstance(dtype, str) else dtype
     dev = torch.device(device)
     # generate directly on the target device: at T=128K per shard the K+V
     # payload is ~2 GB bf16 — a host round-trip (reference model.py:51-53)
@@ -58,7 +62,7 @@ def make_data(
     k = torch.randn((b, hkv, t, d), generator=g, device=gen_dev, dtype=torch.float32)
     v = torch.randn((b, hkv, t, d), generator=g, device=gen_dev, dtype=torch.float32)
     return (
-        q.to(device=dev, dtype=td),
+        q.to(device=dev, dtype=q_td),
         k.to(device=dev, dtype=td),
         v.to(device=dev, dtype=td),
     )
