@@ -198,3 +198,14 @@ def test_fp8_decode_gqa_32k(ext):
 def test_fp8_decode_odd(ext):
     _check_fp8_decode(1, 4, 4, 1000)
     _check_fp8_decode(1, 4, 4, 100)
+
+
+def test_probe_mfma32_layout(ext):
+    """v_mfma_f32_32x32x16_bf16 lane maps: A[row=l&31][k=(l>>5)*8+e],
+    B[k][col=l&31], C[col=l&31][row=(reg&3)+8*(reg>>2)+4*(l>>5)]."""
+    torch.manual_seed(7)
+    a = (torch.randn(32, 16) * 2).bfloat16().cuda()
+    b = (torch.arange(16 * 32).reshape(16, 32).float() % 5 - 2).bfloat16().cuda()
+    c = ext.probe_mfma32(a, b)
+    ref = a.float() @ b.float()
+    torch.testing.assert_close(c.cpu(), ref.cpu(), rtol=1e-2, atol=1e-2)
