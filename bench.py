@@ -51,6 +51,8 @@ def parse_args(argv=None):
     p.add_argument("--q-len", type=int, default=1)
     p.add_argument("--causal", action="store_true")
     p.add_argument("--combine", type=str, default="auto")
+    p.add_argument("--profile", type=str, default=None, metavar="DIR",
+                   help="export a torch.profiler chrome trace of a few steps")
     return p.parse_args(argv)
 
 
@@ -77,6 +79,16 @@ def run(rank: int, world: int, args) -> None:
 
         for _ in range(args.warmup):
             step()
+        if args.profile and rank == 0:
+            # tracing subsystem: kernel-level chrome trace
+            from torch.profiler import ProfilerActivity, profile
+
+            with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
+                for _ in range(3):
+                    step()
+                torch.cuda.synchronize()
+            os.makedirs(args.profile, exist_ok=True)
+            prof.export_chrome_trace(os.path.join(args.profile, "bench_trace.json"))
         with StepTimer(device) as t:
             for _ in range(args.steps):
                 step()

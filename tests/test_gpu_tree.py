@@ -37,3 +37,32 @@ def test_gqa_tree_attention_gpu():
     out = ta.tree_attention(q, k, v)
     ref = ta.attention_reference(q.cpu(), k.cpu(), v.cpu())
     torch.testing.assert_close(out.cpu(), ref, rtol=2.5e-2, atol=2.5e-2)
+
+
+def test_fp8_gqa_tree_attention_gpu():
+    """BASELINE config 5 path end-to-end: fp8 KV + GQA through tree_attention."""
+    import tree_attention_torch_amd as ta
+
+    torch.manual_seed(2)
+    q, k, v = ta.make_data(
+        (1, 32, 8192, 128), rank=0, device="cuda", dtype="fp8", kv_heads=4
+    )
+    out = ta.tree_attention(q, k, v)
+    ref = ta.attention_reference(q.cpu(), k.cpu().float(), v.cpu().float())
+    torch.testing.assert_close(out.cpu(), ref, rtol=0.15, atol=0.15)
+
+
+def test_chunked_prefill_tree_attention_gpu():
+    """Chunked causal prefill through tree_attention (single rank): the
+    q_chunk path must agree with one-shot prefill and the oracle."""
+    import tree_attention_torch_amd as ta
+
+    torch.manual_seed(3)
+    tq = 2048
+    q, k, v = ta.make_data((1, 8, tq, 128), rank=0, device="cuda",
+                           q_len=tq, dtype="bf16")
+    out_chunked = ta.tree_attention(q, k, v, is_causal=True, q_chunk=512)
+    out_oneshot = ta.tree_attention(q, k, v, is_causal=True, q_chunk=tq)
+    torch.testing.assert_close(out_chunked, out_oneshot, rtol=1e-4, atol=1e-4)
+    ref = ta.attention_reference(q.cpu(), k.cpu(), v.cpu(), is_causal=True)
+    torch.testing.assert_close(out_chunked.cpu(), ref, rtol=3e-2, atol=3e-2)
