@@ -209,3 +209,29 @@ def test_probe_mfma32_layout(ext):
     c = ext.probe_mfma32(a, b)
     ref = a.float() @ b.float()
     torch.testing.assert_close(c.cpu(), ref.cpu(), rtol=1e-2, atol=1e-2)
+
+
+def _check_fp16(b, hq, hkv, t, tq=1, causal=False, tol=2.5e-2):
+    from tree_attention_torch_amd.ops.flash import local_attention
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+
+    torch.manual_seed(11)
+    q = torch.randn(b, hq, tq, 128, device="cuda").half()
+    k = torch.randn(b, hkv, t, 128, device="cuda").half()
+    v = torch.randn(b, hkv, t, 128, device="cuda").half()
+    q_off = t - tq
+    out, lse = local_attention(q, k, v, is_causal=causal, q_offset=q_off)
+    ref_out, ref_lse = flash_res_lse(q.cpu(), k.cpu(), v.cpu(),
+                                     is_causal=causal, q_offset=q_off)
+    torch.testing.assert_close(out.cpu(), ref_out, rtol=tol, atol=tol)
+    torch.testing.assert_close(lse.cpu(), ref_lse, rtol=1e-3, atol=1e-3)
+
+
+def test_fp16_decode(ext):
+    """Reference dtype parity: model.py used .half() (fp16) tensors."""
+    _check_fp16(1, 4, 4, 2048)
+    _check_fp16(1, 8, 2, 1000)  # GQA + tail
+
+
+def test_fp16_prefill(ext):
+    _check_fp16(1, 2, 2, 512, tq=512, causal=True)
