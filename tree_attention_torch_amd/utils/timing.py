@@ -43,7 +43,10 @@ class StepTimer:
 
     def max_over_ranks(self) -> float:
         if dist.is_available() and dist.is_initialized():
-            t = torch.tensor([self.elapsed], dtype=torch.float64)
+            # NCCL/RCCL reduces CUDA tensors only; gloo wants CPU
+            dev = "cuda" if (torch.cuda.is_available()
+                             and dist.get_backend() == "nccl") else "cpu"
+            t = torch.tensor([self.elapsed], dtype=torch.float64, device=dev)
             dist.all_reduce(t, op=dist.ReduceOp.MAX)
             return float(t.item())
         return self.elapsed
