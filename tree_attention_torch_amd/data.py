@@ -56,16 +56,20 @@ def make_data(
     # would serialize ranks behind PCIe for no reason.
     gen_dev = dev if dev.type == "cuda" else torch.device("cpu")
     g = torch.Generator(device=gen_dev)
+    # draw directly in a 2-byte dtype (fp8 can't be drawn: go through bf16):
+    # an fp32 staging draw would need 2x the KV bytes — at HBM-cap sequences
+    # (tens of GB of KV per GPU) that OOMs long before the cache does.
+    draw_td = td if td in (torch.bfloat16, torch.float16, torch.float32) \
+        else torch.bfloat16
     g.manual_seed(seed)
-    q = torch.randn((b, h, q_len, d), generator=g, device=gen_dev, dtype=torch.float32)
+    q = torch.randn((b, h, q_len, d), generator=g, device=gen_dev, dtype=q_td)
     g.manual_seed(seed + 1 + rank)
-    k = torch.randn((b, hkv, t, d), generator=g, device=gen_dev, dtype=torch.float32)
-    v = torch.randn((b, hkv, t, d), generator=g, device=gen_dev, dtype=torch.float32)
-    return (
-        q.to(device=dev, dtype=q_td),
-        k.to(device=dev, dtype=td),
-        v.to(device=dev, dtype=td),
-    )
+    k = torch.randn((b, hkv, t, d), generator=g, device=gen_dev, dtype=draw_td)
+    v = torch.randn((b, hkv, t, d), generator=g, device=gen_dev, dtype=draw_td)
+    if k.dtype != td:
+        k = k.to(td)
+        v = v.to(td)
+    return (q.to(dev), k.to(dev), v.to(dev))
 
 
 __all__ = ["make_data"]
