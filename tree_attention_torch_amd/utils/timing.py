@@ -16,11 +16,12 @@ import torch.distributed as dist
 
 
 def barrier_sync(device: torch.device | None = None) -> None:
-    if torch.cuda.is_available():
+    sync = torch.cuda.is_available() and (device is None or device.type == "cuda")
+    if sync:
         torch.cuda.synchronize(device)
     if dist.is_available() and dist.is_initialized():
         dist.barrier()
-    if torch.cuda.is_available():
+    if sync:
         torch.cuda.synchronize(device)
 
 
