@@ -1,4 +1,9 @@
-"""Sanity: the RCCL tree-combine path with 2 ranks sharing one GPU."""
+"""Sanity: the RCCL tree-combine path with 2 ranks sharing one GPU.
+
+NOTE: RCCL refuses two ranks on one device ("Duplicate GPU detected"), so
+this cannot pass on a 1-GPU box — kept as the ready-made check for any
+multi-GPU box. The collective logic itself is covered by the gloo
+world_size 2/4 tests in tests/test_sharded_cpu.py."""
 import os, sys, torch
 import torch.multiprocessing as mp
 
