@@ -31,11 +31,13 @@ from .parallel.combine import (
 )
 from .parallel.pg import cleanup, is_distributed, local_device, setup
 from .parallel.tree import TreeAttention, tree_attention, tree_decode
+from .session import DecodeSession
 from .utils.logging import logger
 
 __version__ = "0.1.0"
 
 __all__ = [
+    "DecodeSession",
     "TreeAttention",
     "TreeAttentionConfig",
     "attention_reference",
