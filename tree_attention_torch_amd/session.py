@@ -62,7 +62,8 @@ class DecodeSession:
         self.group = group
         self.block = block
         td = _DTYPES[kv_dtype] if isinstance(kv_dtype, str) else kv_dtype
-        local_cap = (max_tokens // (block * self.world) + 2) * block
+        blocks_total = (max_tokens + block - 1) // block + 1
+        local_cap = ((blocks_total + self.world - 1) // self.world) * block
         self.k = torch.empty(batch, kv_heads, local_cap, head_dim,
                              dtype=td, device=self.device)
         self.v = torch.empty_like(self.k)
