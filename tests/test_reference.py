@@ -126,3 +126,16 @@ def test_half_precision_inputs_close_to_fp32():
     out32 = attention_reference(q, k, v)
     out16 = attention_reference(q.bfloat16(), k.bfloat16(), v.bfloat16())
     torch.testing.assert_close(out16, out32, rtol=2e-2, atol=2e-2)
+
+
+def test_driver_main_cpu(monkeypatch, tmp_path):
+    """Reference-parity driver main() runs end-to-end on CPU (model.py:129)."""
+    import os
+
+    os.chdir(tmp_path)
+    from tree_attention_torch_amd.config import TreeAttentionConfig
+    from tree_attention_torch_amd.main import main
+
+    cfg = TreeAttentionConfig(seq_len=256, num_heads=2, head_dim=64,
+                              dtype="fp32", warmup=1, steps=2)
+    main(0, 1, cfg)  # should not raise
