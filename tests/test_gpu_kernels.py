@@ -271,3 +271,8 @@ def test_fuzz_shapes(ext):
         if tq > t:
             tq = t
         _check_decode(b, hkv * g, hkv, t, tq=tq, causal=causal, tol=3e-2)
+
+
+def test_fp8_prefill(ext):
+    _check_fp8_decode(1, 2, 2, 512, tq=512, causal=True, tol=0.15)
+    _check_fp8_decode(1, 4, 1, 300, tq=300, causal=True, tol=0.15)  # GQA+odd
