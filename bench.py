@@ -98,6 +98,9 @@ def run(rank: int, world: int, args) -> None:
             seq_total = args.seq_per_gpu * world
             tokens = args.batch * seq_total * args.steps
             value = tokens / elapsed
+            kv_bytes_per_tok = ((args.kv_heads or args.heads) * args.head_dim
+                                * 2 * (1 if args.dtype == "fp8" else 2))
+            eff_tbps = value * kv_bytes_per_tok / 1e12
             result = {
                 "metric": "attention tokens/sec",
                 "value": value,
@@ -122,6 +125,7 @@ def run(rank: int, world: int, args) -> None:
                     "q_len": args.q_len,
                     "parallelism": f"sp{world}",
                     "combine": args.combine,
+                    "effective_kv_tbps_per_gpu": round(eff_tbps / world, 3),
                     "device": "MI355X" if on_gpu else "cpu",
                 },
             }

@@ -116,7 +116,7 @@ def tree_combine_allreduce(
 
     if async_op:
         class _Handle:
-            def wait(self_inner):
+            def wait(self):
                 work.wait()
                 return _finish()
 
@@ -153,7 +153,7 @@ def tree_combine_allgather(
 
     if async_op:
         class _Handle:
-            def wait(self_inner):
+            def wait(self):
                 work.wait()
                 return _finish()
 
@@ -179,7 +179,7 @@ def tree_combine(
     if not (dist.is_available() and dist.is_initialized()) or dist.get_world_size(group) == 1:
         if async_op:
             class _Handle:
-                def wait(self_inner):
+                def wait(self):
                     return out.float(), lse.float()
 
             return _Handle()
