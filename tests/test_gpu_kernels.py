@@ -276,3 +276,21 @@ def test_fuzz_shapes(ext):
 def test_fp8_prefill(ext):
     _check_fp8_decode(1, 2, 2, 512, tq=512, causal=True, tol=0.15)
     _check_fp8_decode(1, 4, 1, 300, tq=300, causal=True, tol=0.15)  # GQA+odd
+
+
+def test_head_dim_64_padding_path(ext):
+    """Narrow heads route through the exact zero-padding fallback."""
+    from tree_attention_torch_amd.ops.flash import local_attention
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+
+    torch.manual_seed(13)
+    for d, tq in ((64, 1), (64, 256), (96, 1)):
+        q = torch.randn(1, 4, tq, d, device="cuda").bfloat16()
+        k = torch.randn(1, 4, 512, d, device="cuda").bfloat16()
+        v = torch.randn(1, 4, 512, d, device="cuda").bfloat16()
+        out, lse = local_attention(q, k, v, is_causal=(tq > 1),
+                                   q_offset=512 - tq)
+        ref_out, ref_lse = flash_res_lse(q.cpu(), k.cpu(), v.cpu(),
+                                         is_causal=(tq > 1), q_offset=512 - tq)
+        torch.testing.assert_close(out.cpu(), ref_out, rtol=2.5e-2, atol=2.5e-2)
+        torch.testing.assert_close(lse.cpu(), ref_lse, rtol=1e-3, atol=1e-3)

@@ -72,6 +72,23 @@ def local_attention(
                 "`python setup.py build_ext --inplace` (or __graft_entry__.build()). "
                 f"Import error: {_EXT_ERR}"
             )
+        d = q.shape[-1]
+        if d < 128:
+            # exact zero-padding fallback for narrow heads: padded dims add 0
+            # to every q.k score and the padded output columns are sliced off.
+            # (softmax_scale above was computed from the REAL D.) Native
+            # D<128 kernels are future work; this costs the padding copies.
+            import torch.nn.functional as F
+
+            pad = 128 - d
+            out, lse = ext.flash_attention(
+                F.pad(q, (0, pad)).contiguous(),
+                F.pad(k, (0, pad)).contiguous(),
+                F.pad(v, (0, pad)).contiguous(),
+                float(softmax_scale), bool(is_causal), int(q_offset),
+                int(kv_offset),
+            )
+            return out[..., :d].contiguous(), lse
         return ext.flash_attention(
             q.contiguous(),
             k.contiguous(),
