@@ -51,12 +51,19 @@ def setup(
     os.environ.setdefault("MASTER_ADDR", master_addr)
     os.environ.setdefault("MASTER_PORT", str(master_port))
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # TREE_ATTN_BACKEND=gloo lets multi-rank GPU code paths run with
+        # several ranks sharing one device (RCCL refuses duplicate devices)
+        backend = os.environ.get(
+            "TREE_ATTN_BACKEND",
+            "nccl" if torch.cuda.is_available() else "gloo",
+        )
     kwargs = {}
     if backend == "nccl":
         local = int(os.environ.get("LOCAL_RANK", rank))
         torch.cuda.set_device(local)
         kwargs["device_id"] = torch.device(f"cuda:{local}")
+    elif backend == "gloo" and torch.cuda.is_available():
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
     dist.init_process_group(
         backend,
         rank=rank,
