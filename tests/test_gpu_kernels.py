@@ -294,3 +294,21 @@ def test_head_dim_64_padding_path(ext):
                                          is_causal=(tq > 1), q_offset=512 - tq)
         torch.testing.assert_close(out.cpu(), ref_out, rtol=2.5e-2, atol=2.5e-2)
         torch.testing.assert_close(lse.cpu(), ref_lse, rtol=1e-3, atol=1e-3)
+
+
+def test_head_dim_64_native_decode(ext):
+    """D=64 decode runs the native 128-B-row kernel (no padding)."""
+    from tree_attention_torch_amd.ops.flash import local_attention
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+
+    torch.manual_seed(14)
+    for b, hq, hkv, t, dt in ((1, 8, 8, 4096, torch.bfloat16),
+                              (1, 32, 4, 1000, torch.bfloat16),
+                              (2, 4, 4, 300, torch.float16)):
+        q = torch.randn(b, hq, 1, 64, device="cuda").to(dt)
+        k = torch.randn(b, hkv, t, 64, device="cuda").to(dt)
+        v = torch.randn(b, hkv, t, 64, device="cuda").to(dt)
+        out, lse = local_attention(q, k, v)
+        ref_out, ref_lse = flash_res_lse(q.cpu(), k.cpu(), v.cpu())
+        torch.testing.assert_close(out.cpu(), ref_out, rtol=2.5e-2, atol=2.5e-2)
+        torch.testing.assert_close(lse.cpu(), ref_lse, rtol=1e-3, atol=1e-3)
