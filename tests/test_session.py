@@ -99,3 +99,20 @@ def test_session_graphed_decode():
         ref, _ = flash_res_lse(q.cpu(), ks[:, :, : t + 1].cpu(),
                                vs[:, :, : t + 1].cpu())
         torch.testing.assert_close(out.cpu(), ref, rtol=2.5e-2, atol=2.5e-2)
+
+
+@pytest.mark.gpu
+def test_session_head_dim_64():
+    """Non-128 head dims fall back from the cache binding to the generic
+    (native-D64 / padded) path."""
+    torch.manual_seed(4)
+    b, h, d = 1, 4, 64
+    sess = DecodeSession(b, h, d, max_tokens=1024, device="cuda",
+                         kv_dtype="bf16", block=128)
+    ks = torch.randn(b, h, 600, d, device="cuda").bfloat16()
+    vs = torch.randn(b, h, 600, d, device="cuda").bfloat16()
+    sess.prefill(ks, vs)
+    q = torch.randn(b, h, 1, d, device="cuda").bfloat16()
+    out = sess.attend(q)
+    ref, _ = flash_res_lse(q.cpu(), ks.cpu(), vs.cpu())
+    torch.testing.assert_close(out.cpu(), ref, rtol=2.5e-2, atol=2.5e-2)

@@ -86,10 +86,15 @@ def local_attention(
             import torch.nn.functional as F
 
             pad = 128 - d
+
+            def _pad(t):
+                if t.dtype == torch.float8_e4m3fn:  # F.pad lacks fp8 support
+                    return F.pad(t.view(torch.uint8), (0, pad)).view(
+                        torch.float8_e4m3fn).contiguous()
+                return F.pad(t, (0, pad)).contiguous()
+
             out, lse = ext.flash_attention(
-                F.pad(q, (0, pad)).contiguous(),
-                F.pad(k, (0, pad)).contiguous(),
-                F.pad(v, (0, pad)).contiguous(),
+                _pad(q), _pad(k), _pad(v),
                 float(softmax_scale), bool(is_causal), int(q_offset),
                 int(kv_offset),
             )

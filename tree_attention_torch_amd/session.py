@@ -85,7 +85,9 @@ class DecodeSession:
         self.total += 1
 
     def _local_partial(self, q: torch.Tensor, softmax_scale: float | None):
-        if q.device.type == "cuda":
+        if q.device.type == "cuda" and self.k.size(3) == 128:
+            # the zero-copy cache binding is head_dim-128 native; other dims
+            # take the generic path below (padding inside local_attention)
             # zero-copy cache path: the kernel takes the cache's head stride
             # and reads the LIVE length from a device scalar, so this call
             # is hipGraph-capturable (graphed_attend) and never copies KV.
