@@ -312,3 +312,21 @@ def test_head_dim_64_native_decode(ext):
         ref_out, ref_lse = flash_res_lse(q.cpu(), k.cpu(), v.cpu())
         torch.testing.assert_close(out.cpu(), ref_out, rtol=2.5e-2, atol=2.5e-2)
         torch.testing.assert_close(lse.cpu(), ref_lse, rtol=1e-3, atol=1e-3)
+
+
+def test_head_dim_64_native_prefill(ext):
+    from tree_attention_torch_amd.ops.flash import local_attention
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+
+    torch.manual_seed(15)
+    for b, h, t, tq, dt in ((1, 4, 512, 512, torch.bfloat16),
+                            (1, 2, 300, 300, torch.float16),
+                            (1, 8, 1024, 256, torch.bfloat16)):
+        q = torch.randn(b, h, tq, 64, device="cuda").to(dt)
+        k = torch.randn(b, h, t, 64, device="cuda").to(dt)
+        v = torch.randn(b, h, t, 64, device="cuda").to(dt)
+        out, lse = local_attention(q, k, v, is_causal=True, q_offset=t - tq)
+        ref_out, ref_lse = flash_res_lse(q.cpu(), k.cpu(), v.cpu(),
+                                         is_causal=True, q_offset=t - tq)
+        torch.testing.assert_close(out.cpu(), ref_out, rtol=2.5e-2, atol=2.5e-2)
+        torch.testing.assert_close(lse.cpu(), ref_lse, rtol=1e-3, atol=1e-3)

@@ -73,11 +73,8 @@ def local_attention(
                 f"Import error: {_EXT_ERR}"
             )
         d = q.shape[-1]
-        native_d64 = (
-            d == 64
-            and q.shape[2] * (q.shape[1] // k.shape[1]) <= 16  # decode shape
-            and k.dtype in (torch.bfloat16, torch.float16)
-        )
+        # D=64 is native for bf16/fp16 (decode and prefill kernels)
+        native_d64 = d == 64 and k.dtype in (torch.bfloat16, torch.float16)
         if d < 128 and not native_d64:
             # exact zero-padding fallback for narrow heads: padded dims add 0
             # to every q.k score and the padded output columns are sliced off.
