@@ -116,3 +116,14 @@ def test_session_head_dim_64():
     out = sess.attend(q)
     ref, _ = flash_res_lse(q.cpu(), ks.cpu(), vs.cpu())
     torch.testing.assert_close(out.cpu(), ref, rtol=2.5e-2, atol=2.5e-2)
+
+
+def test_session_capacity_overflow_raises():
+    sess = DecodeSession(1, 2, 16, max_tokens=8, device="cpu",
+                         kv_dtype="fp32", block=4)
+    k1 = torch.randn(1, 2, 1, 16)
+    for _ in range(12):  # local_cap = 3 blocks of 4 = 12
+        sess.append(k1, k1)
+    with pytest.raises(RuntimeError, match="capacity exceeded"):
+        for _ in range(8):
+            sess.append(k1, k1)

@@ -77,6 +77,14 @@ class DecodeSession:
         """Append one token's K/V (B, Hkv, 1, D). Every rank calls this with
         the same tensors; only the owning rank stores them."""
         if self._owner(self.total) == self.rank:
+            if self.local_len >= self.k.size(2):
+                # without this check the slice assignment below would be an
+                # EMPTY slice and a size-1 source legally expands to 0 rows:
+                # the append would silently drop the token
+                raise RuntimeError(
+                    f"DecodeSession capacity exceeded: local shard full at "
+                    f"{self.local_len} tokens (grow max_tokens)"
+                )
             self.k[:, :, self.local_len : self.local_len + 1] = \
                 k_new.to(self.k.dtype)
             self.v[:, :, self.local_len : self.local_len + 1] = \
@@ -176,6 +184,11 @@ class DecodeSession:
         while t < t_total:
             n = min(self.block - self.total % self.block, t_total - t)
             if self._owner(self.total) == self.rank:
+                if self.local_len + n > self.k.size(2):
+                    raise RuntimeError(
+                        f"DecodeSession capacity exceeded during prefill at "
+                        f"{self.local_len}+{n} tokens (grow max_tokens)"
+                    )
                 self.k[:, :, self.local_len : self.local_len + n] = \
                     k_seq[:, :, t : t + n].to(self.k.dtype)
                 self.v[:, :, self.local_len : self.local_len + n] = \
