@@ -95,3 +95,42 @@ def test_make_data_replicates_q_shards_kv():
     torch.testing.assert_close(qa, qb)  # Q replicated
     assert not torch.allclose(ka, kb)  # K sharded via seed
     assert not torch.allclose(va, vb)
+
+
+def _worker_gqa(rank, world_size, port):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from tree_attention_torch_amd.parallel.tree import tree_attention
+
+        torch.manual_seed(7)
+        b, hq, hkv, d = 1, 8, 2, 32
+        t_total, q_len = 64, 24
+        t_local = t_total // world_size
+        q = torch.randn(b, hq, q_len, d)
+        k_full = torch.randn(b, hkv, t_total, d)
+        v_full = torch.randn(b, hkv, t_total, d)
+        k = k_full[..., rank * t_local : (rank + 1) * t_local, :]
+        v = v_full[..., rank * t_local : (rank + 1) * t_local, :]
+        out = tree_attention(q, k, v, is_causal=True, q_chunk=8)
+        ref, _ = flash_res_lse(q, k_full, v_full, is_causal=True,
+                               q_offset=t_total - q_len)
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_gqa_chunked_prefill_ws2():
+    _PORT[0] += 1
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker_gqa, args=(r, 2, _PORT[0]))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=120)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]

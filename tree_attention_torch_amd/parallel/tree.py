@@ -76,7 +76,9 @@ def tree_attention(
     if q_chunk is None:
         # decode and small prefill: single chunk; large prefill: chunk so the
         # packed combine payload stays ~tens of MB and overlap has depth.
-        q_chunk = tq if tq <= 4096 else 4096
+        # With no collective to overlap (world == 1) chunking only adds
+        # launches, so run the whole Tq in one kernel call.
+        q_chunk = tq if (tq <= 4096 or world == 1) else 4096
 
     if tq <= q_chunk:
         out_l, lse_l = local_attention(
