@@ -74,11 +74,18 @@ def tree_attention(
         softmax_scale = 1.0 / math.sqrt(q.shape[-1])
 
     if q_chunk is None:
-        # decode and small prefill: single chunk; large prefill: chunk so the
-        # packed combine payload stays ~tens of MB and overlap has depth.
         # With no collective to overlap (world == 1) chunking only adds
-        # launches, so run the whole Tq in one kernel call.
-        q_chunk = tq if (tq <= 4096 or world == 1) else 4096
+        # launches and underfills the chip (measured 5.5x at 16K rows,
+        # H=8): run the whole Tq in one kernel call. For world > 1, chunk
+        # for compute/collective overlap but keep each chunk's grid full:
+        # the prefill kernel launches B*Hq*(chunk/256) blocks and the chip
+        # wants >= 512.
+        if world == 1:
+            q_chunk = tq
+        else:
+            b, hq = q.shape[0], q.shape[1]
+            min_chunk = max(4096, (512 * 256) // max(b * hq, 1))
+            q_chunk = tq if tq <= min_chunk else min_chunk
 
     if tq <= q_chunk:
         out_l, lse_l = local_attention(
