@@ -89,7 +89,7 @@ def main():
         q, k, v = make_data((1, 32, t_local, 128), rank, device,
                             q_len=t_local, dtype="bf16")
         dt = timed(lambda: tree_attention(q, k, v, is_causal=True,
-                                          q_chunk=4096, overlap=True),
+                                          overlap=True),
                    device, 5, 2)
         emit(rank, "4_causal_prefill_32kpergpu_bf16", dt, t_local * world,
              {"n_gpus": world, "seq_total": t_local * world, "dtype": "bf16",
