@@ -1,0 +1,27 @@
+"""Same-box A/B: chunked vs unchunked single-GPU causal prefill."""
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+from tree_attention_torch_amd.parallel.tree import tree_attention
+from tree_attention_torch_amd.data import make_data
+
+dev = torch.device("cuda:0")
+
+def bench(h, t, chunk, steps=6, warm=2):
+    q, k, v = make_data((1, h, t, 128), 0, dev, q_len=t, dtype="bf16")
+    f = lambda: tree_attention(q, k, v, is_causal=True, q_chunk=chunk)
+    for _ in range(warm):
+        f()
+    torch.cuda.synchronize(); t0 = time.perf_counter()
+    for _ in range(steps):
+        f()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / steps * 1e3
+
+for h, t in [(32, 32768), (8, 16384), (32, 65536)]:
+    ck = max(4096, (512 * 256) // h)
+    for rep in range(2):
+        un = bench(h, t, t)
+        ch = bench(h, t, ck)
+        print(f"H={h} T={t} rep{rep}: unchunked {un:.2f} ms  chunked({ck}) {ch:.2f} ms",
+              flush=True)
