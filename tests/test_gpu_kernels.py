@@ -330,3 +330,29 @@ def test_head_dim_64_native_prefill(ext):
                                          is_causal=True, q_offset=t - tq)
         torch.testing.assert_close(out.cpu(), ref_out, rtol=2.5e-2, atol=2.5e-2)
         torch.testing.assert_close(lse.cpu(), ref_lse, rtol=1e-3, atol=1e-3)
+
+
+def test_spec_decode_chunked_route(ext):
+    """Tq 17..64 at G*Tq/16 <= 4 chunks routes through the looped split-KV
+    decode kernel (ops/flash.py _dispatch_hip) — boundary shapes."""
+    _check_decode(1, 4, 4, 2048, tq=17, causal=True)   # first chunked size
+    _check_decode(1, 4, 4, 2048, tq=64, causal=True)   # last (4 chunks)
+    _check_decode(1, 4, 4, 1024, tq=64, causal=False)
+    _check_decode(1, 8, 4, 1024, tq=24, causal=True)   # G=2: tq_per=8, 3 chunks
+
+
+def test_spec_decode_vs_prefill_route_equal(ext):
+    """The two routes must agree: force prefill by making the batch big
+    enough to fill the grid (prefill_blocks >= 512 disables the loop)."""
+    from tree_attention_torch_amd.ops.flash import local_attention
+
+    torch.manual_seed(3)
+    t, tq = 512, 32
+    q = torch.randn(1, 8, tq, 128, device="cuda").bfloat16()
+    k = torch.randn(1, 8, t, 128, device="cuda").bfloat16()
+    v = torch.randn(1, 8, t, 128, device="cuda").bfloat16()
+    out_a, lse_a = local_attention(q, k, v, is_causal=True, q_offset=t - tq)
+    # same math through the prefill kernel route
+    out_b, lse_b = ext.flash_attention(q, k, v, 128 ** -0.5, True, t - tq, 0)
+    torch.testing.assert_close(out_a, out_b, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(lse_a, lse_b, rtol=1e-3, atol=1e-3)

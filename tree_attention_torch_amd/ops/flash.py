@@ -43,6 +43,41 @@ def hip_available() -> bool:
     return _load_extension() is not None
 
 
+def _dispatch_hip(ext, q, k, v, softmax_scale, is_causal, q_offset, kv_offset):
+    """Route between the split-KV decode kernel and the prefill kernel.
+
+    The extension itself routes Tq*G <= 16 to decode (MFMA-M batching) and
+    everything else to prefill (256-row q-blocks). For SMALL query batches
+    just past 16 rows — speculative decode: Tq 17..64 against a long KV —
+    the prefill grid is only B*Hq blocks (vs the 512 the chip wants) with
+    no KV split, so instead loop the decode kernel over <=4 query chunks:
+    each chunk launches the full split-KV grid. Costs n_chunks KV reads,
+    wins whenever the prefill grid would run the chip nearly empty
+    (measured: see profiles/decode_matrix_1gpu.jsonl spec-decode rows).
+    """
+    tq = q.shape[2]
+    g = q.shape[1] // k.shape[1]
+    tq_per = 16 // g
+    n_chunks = -(-tq // tq_per)
+    prefill_blocks = q.shape[0] * q.shape[1] * (-(-tq // 256))
+    if tq_per < tq and n_chunks <= 4 and prefill_blocks < 512:
+        outs, lses = [], []
+        for lo in range(0, tq, tq_per):
+            hi = min(tq, lo + tq_per)
+            o, l = ext.flash_attention(
+                q[..., lo:hi, :].contiguous(), k, v,
+                float(softmax_scale), bool(is_causal),
+                int(q_offset + lo), int(kv_offset),
+            )
+            outs.append(o)
+            lses.append(l)
+        return torch.cat(outs, dim=-2), torch.cat(lses, dim=-1)
+    return ext.flash_attention(
+        q, k, v, float(softmax_scale), bool(is_causal),
+        int(q_offset), int(kv_offset),
+    )
+
+
 def local_attention(
     q: torch.Tensor,
     k: torch.Tensor,
@@ -90,20 +125,14 @@ def local_attention(
                         torch.float8_e4m3fn).contiguous()
                 return F.pad(t, (0, pad)).contiguous()
 
-            out, lse = ext.flash_attention(
-                _pad(q), _pad(k), _pad(v),
-                float(softmax_scale), bool(is_causal), int(q_offset),
-                int(kv_offset),
+            out, lse = _dispatch_hip(
+                ext, _pad(q), _pad(k), _pad(v),
+                softmax_scale, is_causal, q_offset, kv_offset,
             )
             return out[..., :d].contiguous(), lse
-        return ext.flash_attention(
-            q.contiguous(),
-            k.contiguous(),
-            v.contiguous(),
-            float(softmax_scale),
-            bool(is_causal),
-            int(q_offset),
-            int(kv_offset),
+        return _dispatch_hip(
+            ext, q.contiguous(), k.contiguous(), v.contiguous(),
+            softmax_scale, is_causal, q_offset, kv_offset,
         )
     return reference.flash_res_lse(q, k, v, softmax_scale, is_causal, q_offset, kv_offset)
 
