@@ -356,3 +356,9 @@ def test_spec_decode_vs_prefill_route_equal(ext):
     out_b, lse_b = ext.flash_attention(q, k, v, 128 ** -0.5, True, t - tq, 0)
     torch.testing.assert_close(out_a, out_b, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(lse_a, lse_b, rtol=1e-3, atol=1e-3)
+
+
+def test_spec_decode_wide_chunked(ext):
+    """Tq=96 (6 chunks at G=1) stays on the looped decode route when the
+    prefill grid would be underfilled (max_chunks scales with 512/blocks)."""
+    _check_decode(1, 4, 4, 2048, tq=96, causal=True)

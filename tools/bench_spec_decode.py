@@ -20,7 +20,7 @@ def t_ms(f, steps=20, warm=5):
     return (time.perf_counter() - t0) / steps * 1e3
 
 for tkv in (32768, 131072):
-    for tq in (8, 16, 17, 32, 64, 96):
+    for tq in (8, 16, 17, 32, 64, 96, 128, 256):
         q, k, v = make_data((1, 32, tkv, 128), 0, dev, q_len=tq, dtype="bf16")
         scale = 128 ** -0.5
         off = tkv - tq

@@ -60,7 +60,12 @@ def _dispatch_hip(ext, q, k, v, softmax_scale, is_causal, q_offset, kv_offset):
     tq_per = 16 // g
     n_chunks = -(-tq // tq_per)
     prefill_blocks = q.shape[0] * q.shape[1] * (-(-tq // 256))
-    if tq_per < tq and n_chunks <= 4 and prefill_blocks < 512:
+    # Cost model (measured, H=32 @ 32K/128K KV): each decode chunk is one
+    # full-bandwidth KV stream; the underfilled prefill launch costs about
+    # (512/blocks) * 2 streams. Loop while n_chunks is under that.
+    # Tq=17..32: 0.72 vs 9.62 ms at 128K; Tq=96 (6 chunks): 2.0 vs 9.6.
+    max_chunks = max(4, (2 * 512) // max(prefill_blocks, 1))
+    if tq_per < tq and n_chunks <= max_chunks and prefill_blocks < 512:
         outs, lses = [], []
         for lo in range(0, tq, tq_per):
             hi = min(tq, lo + tq_per)
