@@ -60,12 +60,14 @@ def _dispatch_hip(ext, q, k, v, softmax_scale, is_causal, q_offset, kv_offset):
     tq_per = 16 // g
     n_chunks = -(-tq // tq_per)
     prefill_blocks = q.shape[0] * q.shape[1] * (-(-tq // 256))
-    # Cost model (measured, H=32 @ 32K/128K KV): each decode chunk is one
-    # full-bandwidth KV stream; the underfilled prefill launch costs about
-    # (512/blocks) * 2 streams. Loop while n_chunks is under that.
-    # Tq=17..32: 0.72 vs 9.62 ms at 128K; Tq=96 (6 chunks): 2.0 vs 9.6.
-    max_chunks = max(4, (2 * 512) // max(prefill_blocks, 1))
-    if tq_per < tq and n_chunks <= max_chunks and prefill_blocks < 512:
+    # Cost model (measured sweep, H=32 @ 32K/128K KV): each decode chunk is
+    # one full-bandwidth KV stream; the underfilled prefill launch costs
+    # about (512/blocks) streams. Loop while n_chunks is under that.
+    # Tq=17..32: 0.72 vs 9.62 ms at 128K; Tq=128: 2.65 vs 9.62; crossover
+    # measured at Tq≈256 (prefill 1.47 vs loop 1.61 at 32K) = the n_chunks
+    # == 512/blocks point, so strict <.
+    max_chunks = max(4, 512 // max(prefill_blocks, 1))
+    if tq_per < tq and n_chunks < max_chunks and prefill_blocks < 512:
         outs, lses = [], []
         for lo in range(0, tq, tq_per):
             hi = min(tq, lo + tq_per)
