@@ -18,10 +18,13 @@ def bench(h, t, chunk, steps=6, warm=2):
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / steps * 1e3
 
+import sys
+chunks = [int(x) for x in sys.argv[1:]] or None
 for h, t in [(32, 32768), (8, 16384), (32, 65536)]:
-    ck = max(4096, (512 * 256) // h)
+    cks = chunks or [t, max(4096, (512 * 256) // h)]
     for rep in range(2):
-        un = bench(h, t, t)
-        ch = bench(h, t, ck)
-        print(f"H={h} T={t} rep{rep}: unchunked {un:.2f} ms  chunked({ck}) {ch:.2f} ms",
-              flush=True)
+        line = f"H={h} T={t} rep{rep}:"
+        for ck in cks:
+            ms = bench(h, t, min(ck, t))
+            line += f"  chunk{min(ck, t)}={ms:.2f}ms"
+        print(line, flush=True)
