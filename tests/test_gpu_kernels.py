@@ -362,3 +362,20 @@ def test_spec_decode_wide_chunked(ext):
     """Tq=96 (6 chunks at G=1) stays on the looped decode route when the
     prefill grid would be underfilled (max_chunks scales with 512/blocks)."""
     _check_decode(1, 4, 4, 2048, tq=96, causal=True)
+
+
+def test_combine_packed_kernel(ext):
+    """HIP packed-combine (the all-gather epilogue) vs the eager oracle,
+    including a fully-masked shard (lse = -inf)."""
+    from tree_attention_torch_amd.parallel.combine import combine_partials
+
+    torch.manual_seed(11)
+    for s, b, h, tq, d in [(8, 1, 32, 1, 128), (3, 2, 4, 5, 64), (2, 1, 1, 1, 128)]:
+        outs = torch.randn(s, b, h, tq, d, device="cuda")
+        lses = torch.randn(s, b, h, tq, device="cuda") * 4
+        lses[0] = float("-inf")  # rank 0 fully masked
+        packed = torch.cat([outs, lses.unsqueeze(-1)], dim=-1).contiguous()
+        out_k, lse_k = ext.combine_packed(packed.view(-1), s, b, h, tq, d)
+        out_e, lse_e = combine_partials(outs.cpu(), lses.cpu())
+        torch.testing.assert_close(out_k.cpu(), out_e, rtol=1e-5, atol=1e-5)
+        torch.testing.assert_close(lse_k.cpu(), lse_e, rtol=1e-5, atol=1e-5)

@@ -146,6 +146,15 @@ def tree_combine_allgather(
     work = dist.all_gather_into_tensor(gathered, flat, group=group, async_op=async_op)
 
     def _finish():
+        if gathered.is_cuda and d in (64, 128):
+            # one HIP kernel instead of the eager amax/exp/sum/div/log
+            # chain — the per-step epilogue on the N-GPU decode path.
+            from ..ops import flash
+
+            ext = flash._load_extension()
+            if ext is not None:
+                out_g, lse_g = ext.combine_packed(gathered, ws, b, h, tq, d)
+                return out_g, lse_g
         stacked = gathered.view((ws,) + packed.shape)
         outs = stacked[..., :d]
         lses = stacked[..., d]
