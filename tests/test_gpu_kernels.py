@@ -379,3 +379,25 @@ def test_combine_packed_kernel(ext):
         out_e, lse_e = combine_partials(outs.cpu(), lses.cpu())
         torch.testing.assert_close(out_k.cpu(), out_e, rtol=1e-5, atol=1e-5)
         torch.testing.assert_close(lse_k.cpu(), lse_e, rtol=1e-5, atol=1e-5)
+
+
+def test_combine_rescale_finish_kernels(ext):
+    """The fused all-reduce halves reproduce combine_partials: emulate the
+    collective locally (elementwise max of m, sum of packed)."""
+    from tree_attention_torch_amd.parallel.combine import combine_partials
+
+    torch.manual_seed(12)
+    b, h, tq, d = 1, 8, 3, 128
+    outs = torch.randn(2, b, h, tq, d, device="cuda")
+    lses = torch.randn(2, b, h, tq, device="cuda") * 4
+    lses[1, 0, :2] = float("-inf")  # rank 1 partially masked
+    m = torch.where(torch.isfinite(lses), lses,
+                    torch.full_like(lses, -80.0)).amax(0)
+    packed = (ext.combine_rescale_pack(outs[0].contiguous(),
+                                       lses[0].contiguous(), m.contiguous())
+              + ext.combine_rescale_pack(outs[1].contiguous(),
+                                         lses[1].contiguous(), m.contiguous()))
+    out_k, lse_k = ext.combine_finish(packed.contiguous(), m.contiguous())
+    out_e, lse_e = combine_partials(outs.cpu(), lses.cpu())
+    torch.testing.assert_close(out_k.cpu(), out_e, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(lse_k.cpu(), lse_e, rtol=1e-5, atol=1e-5)
