@@ -134,3 +134,43 @@ def test_gqa_chunked_prefill_ws2():
     for p in procs:
         p.join(timeout=120)
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
+def _worker_uneven(rank, world_size, port):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from tree_attention_torch_amd.parallel.tree import tree_attention
+
+        torch.manual_seed(5)
+        b, h, d, t_total, q_len = 1, 4, 32, 96, 8
+        bounds = [0, 40, 96]  # rank 0 holds 40 keys, rank 1 holds 56
+        q = torch.randn(b, h, q_len, d)
+        k_full = torch.randn(b, h, t_total, d)
+        v_full = torch.randn(b, h, t_total, d)
+        lo, hi = bounds[rank], bounds[rank + 1]
+        out = tree_attention(
+            q, k_full[..., lo:hi, :], v_full[..., lo:hi, :],
+            is_causal=True, kv_offset=lo, total_kv=t_total,
+        )
+        ref, _ = flash_res_lse(q, k_full, v_full, is_causal=True,
+                               q_offset=t_total - q_len)
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_uneven_shards_ws2():
+    _PORT[0] += 1
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker_uneven, args=(r, 2, _PORT[0]))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=120)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]

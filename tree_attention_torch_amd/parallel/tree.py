@@ -44,6 +44,8 @@ def tree_attention(
     q_chunk: int | None = None,
     overlap: bool = True,
     return_lse: bool = False,
+    kv_offset: int | None = None,
+    total_kv: int | None = None,
 ):
     """Attention over the KV sequence sharded across ranks.
 
@@ -57,14 +59,22 @@ def tree_attention(
         combine: "auto" | "allgather" | "allreduce" (parallel/combine.py).
         q_chunk: prefill chunk size along Tq (None = pick automatically).
         overlap: overlap chunk i's collective with chunk i+1's kernel.
+        kv_offset / total_kv: global position of this rank's shard and the
+            global sequence length. Default (None) assumes EQUAL shards —
+            rank r holds [r*T_local, (r+1)*T_local). Pass both explicitly
+            for uneven sharding (e.g. ragged serving caches); the combine
+            is shard-size-agnostic, only the causal positions need them.
 
     Returns:
         out (B, Hq, Tq, D) fp32 (and lse (B, Hq, Tq) if return_lse).
     """
     rank, world = _rank_and_world(group)
     t_local = k.shape[-2]
-    kv_offset = rank * t_local
-    total_kv = world * t_local
+    if (kv_offset is None) != (total_kv is None):
+        raise ValueError("pass kv_offset and total_kv together (or neither)")
+    if kv_offset is None:
+        kv_offset = rank * t_local
+        total_kv = world * t_local
     tq = q.shape[-2]
     # queries sit at the END of the global sequence (prefill over the full
     # sequence has tq == total_kv and q_offset 0).
