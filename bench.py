@@ -79,6 +79,17 @@ def run(rank: int, world: int, args) -> None:
 
         for _ in range(args.warmup):
             step()
+        if on_gpu:
+            # DVFS ramp: a few warmup steps (~ms of activity) leave the GPU
+            # at idle clocks and the first timed steps ~15% slow (measured:
+            # cold bench 0.349 ms/step vs 0.318 after a prior run on the
+            # same box). Spin the SAME step untimed for ~2 s wall clock.
+            spin_until = time.perf_counter() + 2.0
+            spins = 0
+            while time.perf_counter() < spin_until and spins < 500:
+                step()
+                spins += 1
+            torch.cuda.synchronize()
         if args.profile and rank == 0:
             # tracing subsystem: kernel-level chrome trace
             from torch.profiler import ProfilerActivity, profile
