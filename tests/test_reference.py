@@ -139,3 +139,26 @@ def test_driver_main_cpu(monkeypatch, tmp_path):
     cfg = TreeAttentionConfig(seq_len=256, num_heads=2, head_dim=64,
                               dtype="fp32", warmup=1, steps=2)
     main(0, 1, cfg)  # should not raise
+
+
+def test_bench_contract_cpu(capsys):
+    """bench.py emits ONE JSON line with the driver-contract fields."""
+    import json
+
+    import bench
+
+    bench.main(["--steps", "2", "--warmup", "1"])
+    out = capsys.readouterr().out.strip().splitlines()
+    assert len(out) == 1, out
+    d = json.loads(out[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, key
+    assert d["metric"] == "attention tokens/sec"
+    assert d["n_gpus"] == 1 and d["steps"] == 2 and d["warmup"] == 1
+    assert d["scaling"] == "weak" and d["data"] == "synthetic"
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    cfg = d["config"]
+    assert cfg["num_heads"] == 32 and cfg["head_dim"] == 128
+    assert cfg["global_batch"] == 1 and cfg["q_len"] == 1
