@@ -1,0 +1,88 @@
+// probe_sched: price the pwg4x64-style MFMA stream on gfx950 silicon.
+// Build: hipcc --offload-arch=gfx950 -O3 -o /tmp/probe_sched tools/probe_sched.hip
+// Run (GPU box): /tmp/probe_sched   -> cyc/MFMA for each variant.
+//
+// M0: bare 32-MFMA phase, 4 rotating f32x16 accumulators (floor).
+// M1: same + exactly 5 hand-placed independent v_fma_f32 fillers per gap
+//     (the guide's budget: one wave/SIMD hides <= 5 single-issue
+//     instructions per v_mfma_f32_32x32x16_bf16).
+// M2: the same multiset (32 builtin MFMAs + 160 fma) in C, compiler-
+//     scheduled — measures what hipcc does with the freedom.
+// Launch: ONE block of 4 waves (one per SIMD of one CU), occupancy 1.
+#include <hip/hip_runtime.h>
+#include <cstdio>
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+__device__ __forceinline__ unsigned long memtime() {
+  unsigned long t;
+  asm volatile("s_memtime %0" : "=s"(t));
+  return t;
+}
+
+template <int MODE>
+__global__ __launch_bounds__(256, 1) void probe_kernel(
+    float* out, unsigned long* cycles, int iters) {
+  f32x16 c0 = {}, c1 = {}, c2 = {}, c3 = {};
+  bf16x8 a, b;
+  float f0 = 1.0f, f1 = 1.0f;
+  const float k = 1.0000001f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) { a[i] = (__bf16)(threadIdx.x + i); b[i] = (__bf16)(i); }
+  asm volatile("s_waitcnt lgkmcnt(0) vmcnt(0)");
+  __syncthreads();
+  const unsigned long t0 = memtime();
+  for (int it = 0; it < iters; ++it) {
+    if (MODE == 0) {
+      asm volatile("v_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3"
+                   : "+v"(c0), "+v"(c1), "+v"(c2), "+v"(c3)
+                   : "v"(a), "v"(b));
+    } else if (MODE == 1) {
+      asm volatile("v_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7"
+                   : "+v"(c0), "+v"(c1), "+v"(c2), "+v"(c3),
+                     "+v"(f0), "+v"(f1)
+                   : "v"(a), "v"(b), "v"(k));
+    } else {
+#pragma unroll
+      for (int i = 0; i < 32; ++i) {
+        f32x16& c = (i % 4 == 0) ? c0 : (i % 4 == 1) ? c1 : (i % 4 == 2) ? c2 : c3;
+        c = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+        for (int f = 0; f < 5; ++f) {
+          float& d = ((i * 5 + f) % 2) ? f1 : f0;
+          d = __builtin_fmaf(d, k, d);
+        }
+      }
+    }
+  }
+  asm volatile("s_nop 11");
+  const unsigned long t1 = memtime();
+  if (threadIdx.x % 64 == 0) cycles[threadIdx.x / 64] = t1 - t0;
+  // keep results alive
+  out[threadIdx.x] = c0[0] + c1[0] + c2[0] + c3[0] + f0 + f1;
+}
+
+template <int MODE>
+double run(int iters) {
+  float* out; unsigned long* cyc;
+  hipMalloc(&out, 256 * sizeof(float));
+  hipMalloc(&cyc, 4 * sizeof(unsigned long));
+  probe_kernel<MODE><<<1, 256>>>(out, cyc, iters);  // warm
+  probe_kernel<MODE><<<1, 256>>>(out, cyc, iters);
+  hipDeviceSynchronize();
+  unsigned long h[4];
+  hipMemcpy(h, cyc, sizeof(h), hipMemcpyDeviceToHost);
+  hipFree(out); hipFree(cyc);
+  unsigned long mx = 0;
+  for (int i = 0; i < 4; ++i) mx = h[i] > mx ? h[i] : mx;
+  return (double)mx / (iters * 32.0);
+}
+
+int main() {
+  const int iters = 2000;
+  printf("M0 bare 32-MFMA phase:        %.2f cyc/MFMA\n", run<0>(iters));
+  printf("M1 hand-placed 5 fillers/gap: %.2f cyc/MFMA\n", run<1>(iters));
+  printf("M2 compiler-scheduled same:   %.2f cyc/MFMA\n", run<2>(iters));
+  return 0;
+}
