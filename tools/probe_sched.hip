@@ -39,7 +39,7 @@ __global__ __launch_bounds__(256, 1) void probe_kernel(
                    : "+v"(c0), "+v"(c1), "+v"(c2), "+v"(c3)
                    : "v"(a), "v"(b));
     } else if (MODE == 1) {
-      asm volatile("v_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7\n\tv_fma_f32 %6, %6, %8, %6\n\tv_fma_f32 %7, %7, %8, %7"
+      asm volatile("v_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5\n\tv_fma_f32 %4, %4, %8, %4\n\tv_fma_f32 %5, %5, %8, %5"
                    : "+v"(c0), "+v"(c1), "+v"(c2), "+v"(c3),
                      "+v"(f0), "+v"(f1)
                    : "v"(a), "v"(b), "v"(k));
