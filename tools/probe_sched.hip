@@ -15,15 +15,9 @@
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
-__device__ __forceinline__ unsigned long memtime() {
-  unsigned long t;
-  asm volatile("s_memtime %0" : "=s"(t));
-  return t;
-}
-
 template <int MODE>
 __global__ __launch_bounds__(256, 1) void probe_kernel(
-    float* out, unsigned long* cycles, int iters) {
+    float* out, int iters) {
   f32x16 c0 = {}, c1 = {}, c2 = {}, c3 = {};
   bf16x8 a, b;
   float f0 = 1.0f, f1 = 1.0f;
@@ -32,7 +26,6 @@ __global__ __launch_bounds__(256, 1) void probe_kernel(
   for (int i = 0; i < 8; ++i) { a[i] = (__bf16)(threadIdx.x + i); b[i] = (__bf16)(i); }
   asm volatile("s_waitcnt lgkmcnt(0) vmcnt(0)");
   __syncthreads();
-  const unsigned long t0 = memtime();
   for (int it = 0; it < iters; ++it) {
     if (MODE == 0) {
       asm volatile("v_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3\n\tv_mfma_f32_32x32x16_bf16 %0, %4, %5, %0\n\tv_mfma_f32_32x32x16_bf16 %1, %4, %5, %1\n\tv_mfma_f32_32x32x16_bf16 %2, %4, %5, %2\n\tv_mfma_f32_32x32x16_bf16 %3, %4, %5, %3"
@@ -57,32 +50,36 @@ __global__ __launch_bounds__(256, 1) void probe_kernel(
     }
   }
   asm volatile("s_nop 11");
-  const unsigned long t1 = memtime();
-  if (threadIdx.x % 64 == 0) cycles[threadIdx.x / 64] = t1 - t0;
   // keep results alive
   out[threadIdx.x] = c0[0] + c1[0] + c2[0] + c3[0] + f0 + f1;
 }
 
 template <int MODE>
 double run(int iters) {
-  float* out; unsigned long* cyc;
-  hipMalloc(&out, 256 * sizeof(float));
-  hipMalloc(&cyc, 4 * sizeof(unsigned long));
-  probe_kernel<MODE><<<1, 256>>>(out, cyc, iters);  // warm
-  probe_kernel<MODE><<<1, 256>>>(out, cyc, iters);
-  hipDeviceSynchronize();
-  unsigned long h[4];
-  hipMemcpy(h, cyc, sizeof(h), hipMemcpyDeviceToHost);
-  hipFree(out); hipFree(cyc);
-  unsigned long mx = 0;
-  for (int i = 0; i < 4; ++i) mx = h[i] > mx ? h[i] : mx;
-  return (double)mx / (iters * 32.0);
+  float* out;
+  (void)hipMalloc(&out, 256 * sizeof(float));
+  hipEvent_t e0, e1;
+  (void)hipEventCreate(&e0);
+  (void)hipEventCreate(&e1);
+  probe_kernel<MODE><<<1, 256>>>(out, iters);  // warm
+  (void)hipDeviceSynchronize();
+  (void)hipEventRecord(e0);
+  probe_kernel<MODE><<<1, 256>>>(out, iters);
+  (void)hipEventRecord(e1);
+  (void)hipDeviceSynchronize();
+  float ms = 0.f;
+  (void)hipEventElapsedTime(&ms, e0, e1);
+  (void)hipFree(out);
+  // ns per MFMA; cyc assumes ~2.4 GHz shader clock
+  return (double)ms * 1e6 / (iters * 32.0);
 }
 
 int main() {
-  const int iters = 2000;
-  printf("M0 bare 32-MFMA phase:        %.2f cyc/MFMA\n", run<0>(iters));
-  printf("M1 hand-placed 5 fillers/gap: %.2f cyc/MFMA\n", run<1>(iters));
-  printf("M2 compiler-scheduled same:   %.2f cyc/MFMA\n", run<2>(iters));
+  const int iters = 20000;
+  const double ghz = 2.4;  // nominal; scale if rocm-smi shows otherwise
+  double m0 = run<0>(iters), m1 = run<1>(iters), m2 = run<2>(iters);
+  printf("M0 bare 32-MFMA phase:        %.2f ns/MFMA  (%.1f cyc @%.1fGHz)\n", m0, m0 * ghz, ghz);
+  printf("M1 hand-placed 5 fillers/gap: %.2f ns/MFMA  (%.1f cyc @%.1fGHz)\n", m1, m1 * ghz, ghz);
+  printf("M2 compiler-scheduled same:   %.2f ns/MFMA  (%.1f cyc @%.1fGHz)\n", m2, m2 * ghz, ghz);
   return 0;
 }
