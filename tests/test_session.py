@@ -127,3 +127,40 @@ def test_session_capacity_overflow_raises():
     with pytest.raises(RuntimeError, match="capacity exceeded"):
         for _ in range(8):
             sess.append(k1, k1)
+
+
+def test_session_batched_cpu():
+    """Batched serving (B=3): the cache and attend are batch-dim clean."""
+    torch.manual_seed(4)
+    b, h, d = 3, 4, 32
+    sess = DecodeSession(b, h, d, max_tokens=256, device="cpu",
+                         kv_dtype="fp32", block=8)
+    ks = torch.randn(b, h, 33, d)
+    vs = torch.randn(b, h, 33, d)
+    sess.prefill(ks[:, :, :9], vs[:, :, :9])
+    for t in range(9, 33):
+        sess.append(ks[:, :, t : t + 1], vs[:, :, t : t + 1])
+        q = torch.randn(b, h, 1, d)
+        out = sess.attend(q)
+        ref, _ = flash_res_lse(q, ks[:, :, : t + 1], vs[:, :, : t + 1])
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_session_batched_gpu():
+    """Batched serving on the zero-copy cache kernel path (B=4, bf16)."""
+    torch.manual_seed(5)
+    b, h, d = 4, 8, 128
+    sess = DecodeSession(b, h, d, max_tokens=4096, device="cuda",
+                         kv_dtype="bf16", block=256)
+    ks = torch.randn(b, h, 600, d, device="cuda").bfloat16()
+    vs = torch.randn(b, h, 600, d, device="cuda").bfloat16()
+    sess.prefill(ks[:, :, :512], vs[:, :, :512])
+    for t in range(512, 600, 17):
+        q = torch.randn(b, h, 1, d, device="cuda").bfloat16()
+        out = sess.attend(q)
+        ref, _ = flash_res_lse(q.cpu(), ks[:, :, :sess.total].cpu(),
+                               vs[:, :, :sess.total].cpu())
+        torch.testing.assert_close(out.cpu(), ref, rtol=2.5e-2, atol=2.5e-2)
+        for u in range(t, min(t + 17, 600)):
+            sess.append(ks[:, :, u : u + 1], vs[:, :, u : u + 1])
