@@ -19,6 +19,13 @@ template <int MODE>
 __global__ __launch_bounds__(256, 1) void probe_kernel(
     float* out, int iters) {
   f32x16 c0 = {}, c1 = {}, c2 = {}, c3 = {};
+  __shared__ __attribute__((aligned(16))) unsigned short lds[4096];
+  for (int i = threadIdx.x; i < 4096; i += 256) lds[i] = (unsigned short)i;
+  typedef __attribute__((ext_vector_type(2))) unsigned u32x2;
+  u32x2 trd = {0u, 0u};
+  const unsigned lds_addr =
+      (unsigned)(uintptr_t)(__attribute__((address_space(3))) unsigned short*)
+          &lds[(threadIdx.x & 63) * 4];
   bf16x8 a, b;
   float f0 = 1.0f, f1 = 1.0f;
   const float k = 1.0000001f;
@@ -36,6 +43,11 @@ __global__ __launch_bounds__(256, 1) void probe_kernel(
                    : "+v"(c0), "+v"(c1), "+v"(c2), "+v"(c3),
                      "+v"(f0), "+v"(f1)
                    : "v"(a), "v"(b), "v"(k));
+    } else if (MODE == 3) {
+      asm volatile("v_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %0, %6, %7, %0\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %1, %6, %7, %1\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %2, %6, %7, %2\n\tv_exp_f32 %4, %4\n\tv_fma_f32 %4, %4, %8, %4\n\tv_max3_f32 %4, %4, %8, %8\n\tv_cvt_pk_bf16_f32 %4, %4, %8\n\tds_read_b64_tr_b16 %10, %9\n\tv_mfma_f32_32x32x16_bf16 %3, %6, %7, %3\n\tv_exp_f32 %5, %5\n\tv_fma_f32 %5, %5, %8, %5\n\tv_max3_f32 %5, %5, %8, %8\n\tv_cvt_pk_bf16_f32 %5, %5, %8\n\tds_read_b64_tr_b16 %10, %9\n\ts_waitcnt lgkmcnt(0)"
+                   : "+v"(c0), "+v"(c1), "+v"(c2), "+v"(c3),
+                     "+v"(f0), "+v"(f1)
+                   : "v"(a), "v"(b), "v"(k), "v"(lds_addr), "v"(trd));
     } else {
 #pragma unroll
       for (int i = 0; i < 32; ++i) {
@@ -51,7 +63,7 @@ __global__ __launch_bounds__(256, 1) void probe_kernel(
   }
   asm volatile("s_nop 11");
   // keep results alive
-  out[threadIdx.x] = c0[0] + c1[0] + c2[0] + c3[0] + f0 + f1;
+  out[threadIdx.x] = c0[0] + c1[0] + c2[0] + c3[0] + f0 + f1 + (float)trd[0];
 }
 
 template <int MODE>
@@ -81,5 +93,7 @@ int main() {
   printf("M0 bare 32-MFMA phase:        %.2f ns/MFMA  (%.1f cyc @%.1fGHz)\n", m0, m0 * ghz, ghz);
   printf("M1 hand-placed 5 fillers/gap: %.2f ns/MFMA  (%.1f cyc @%.1fGHz)\n", m1, m1 * ghz, ghz);
   printf("M2 compiler-scheduled same:   %.2f ns/MFMA  (%.1f cyc @%.1fGHz)\n", m2, m2 * ghz, ghz);
+  double m3 = run<3>(iters);
+  printf("M3 real filler mix hand-placed (exp/fma/max3/cvt_pk/tr16): %.2f ns/MFMA  (%.1f cyc @%.1fGHz)\n", m3, m3 * ghz, ghz);
   return 0;
 }
