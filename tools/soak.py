@@ -18,7 +18,7 @@ q1, k1, v1 = make_data((1, 32, 131072, 128), 0, dev, dtype="bf16")
 q2, k2, v2 = make_data((1, 32, 524288, 128), 0, dev, dtype="fp8", kv_heads=4)
 qs, ks, vs = make_data((1, 8, 8192, 128), 0, dev, q_len=32, dtype="bf16")
 qp, kp, vp = make_data((1, 8, 8192, 128), 0, dev, q_len=8192, dtype="bf16")
-sess = DecodeSession(1, 8, 128, max_tokens=65536, device=dev, kv_dtype="bf16")
+sess = DecodeSession(1, 8, 128, max_tokens=1 << 21, device=dev, kv_dtype="bf16")
 sess.prefill(ks, vs)
 qg = torch.randn(1, 8, 1, 128, device=dev).bfloat16()
 replay, out_g = sess.graphed_attend(qg)
@@ -51,6 +51,8 @@ while time.time() - t0 < secs:
     assert torch.isfinite(o1).all() and torch.isfinite(o2).all() \
         and torch.isfinite(o3).all()
     it += 1
+    if it % 20000 == 0:
+        print(f"  t={time.time()-t0:.0f}s it={it} len={sess.total}", flush=True)
 torch.cuda.synchronize()
 print(f"SOAK OK: {it} iterations, {checks} oracle spot-checks, "
       f"{time.time()-t0:.0f}s, session len {sess.total}", flush=True)
