@@ -433,3 +433,22 @@ def test_experimental_prefill_variants(ext, flag):
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, env=env, timeout=300)
     assert r.returncode == 0 and "VARIANT_OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_probe_gen6_qkt_stream(ext):
+    """Generated hand-placed QKT_PAIR asm stream (round-2 groundwork):
+    exact vs the builtin-MFMA control and the eager reference."""
+    torch.manual_seed(0)
+    q = (torch.randn(64, 128) * 0.5).bfloat16().cuda()
+    k = (torch.randn(32, 128) * 0.5).bfloat16().cuda()
+    out = ext.probe_gen6_qkt(q, k).cpu()  # (4,64,16): asm j0/j1, builtin j0/j1
+    torch.testing.assert_close(out[0], out[2], rtol=0, atol=0)  # asm == builtin
+    torch.testing.assert_close(out[1], out[3], rtol=0, atol=0)
+    ref = (k.float() @ q.float().T).cpu()
+    lanes = torch.arange(64)
+    row32, h = lanes % 32, lanes // 32
+    for j in range(2):
+        for reg in range(16):
+            key = (reg % 4) + 8 * (reg // 4)
+            exp = ref[key + 4 * h, j * 32 + row32]
+            torch.testing.assert_close(out[j][:, reg], exp, rtol=2e-2, atol=2e-2)
