@@ -162,3 +162,27 @@ def test_bench_contract_cpu(capsys):
     cfg = d["config"]
     assert cfg["num_heads"] == 32 and cfg["head_dim"] == 128
     assert cfg["global_batch"] == 1 and cfg["q_len"] == 1
+
+
+def test_tree_attention_qlen_exceeds_kv_raises():
+    from tree_attention_torch_amd.parallel.tree import tree_attention
+
+    q = torch.randn(1, 2, 16, 32)
+    k = torch.randn(1, 2, 8, 32)
+    v = torch.randn(1, 2, 8, 32)
+    with pytest.raises(ValueError, match="exceeds the global KV length"):
+        tree_attention(q, k, v, is_causal=True)
+
+
+def test_gqa_group_16_oracle():
+    """Largest GQA group the decode kernel supports (G=16)."""
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+
+    torch.manual_seed(9)
+    q = torch.randn(1, 16, 1, 32)
+    k = torch.randn(1, 1, 64, 32)
+    v = torch.randn(1, 1, 64, 32)
+    out, lse = flash_res_lse(q, k, v)
+    ref, _ = flash_res_lse(q, k.expand(1, 16, 64, 32).contiguous(),
+                           v.expand(1, 16, 64, 32).contiguous())
+    torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)

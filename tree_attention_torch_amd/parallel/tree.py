@@ -76,6 +76,12 @@ def tree_attention(
         kv_offset = rank * t_local
         total_kv = world * t_local
     tq = q.shape[-2]
+    if tq > total_kv:
+        raise ValueError(
+            f"tree_attention: q_len {tq} exceeds the global KV length "
+            f"{total_kv} — queries are the LAST Tq positions of the global "
+            "sequence (decode semantics), so each query needs a KV position"
+        )
     # queries sit at the END of the global sequence (prefill over the full
     # sequence has tq == total_kv and q_offset 0).
     q_offset = total_kv - tq
