@@ -452,3 +452,26 @@ def test_probe_gen6_qkt_stream(ext):
             key = (reg % 4) + 8 * (reg // 4)
             exp = ref[key + 4 * h, j * 32 + row32]
             torch.testing.assert_close(out[j][:, reg], exp, rtol=2e-2, atol=2e-2)
+
+
+def test_decode_gqa16(ext):
+    """Largest supported GQA group (G=16) through the MFMA-M batching."""
+    _check_decode(1, 16, 1, 4096)
+    _check_decode(1, 16, 1, 1000, causal=True)
+
+
+def test_fp8_narrow_head_padding(ext):
+    """fp8 KV with d<128 routes through the uint8-view zero-pad fallback."""
+    from tree_attention_torch_amd.ops.flash import local_attention
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+
+    torch.manual_seed(17)
+    d = 96
+    q = torch.randn(1, 4, 1, d, device="cuda").bfloat16()
+    k = (torch.randn(1, 4, 512, d, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+    v = (torch.randn(1, 4, 512, d, device="cuda") * 0.5).to(torch.float8_e4m3fn)
+    out, lse = local_attention(q, k, v, q_offset=511)
+    ref_out, ref_lse = flash_res_lse(
+        q.cpu().float(), k.cpu().float(), v.cpu().float(), q_offset=511)
+    torch.testing.assert_close(out.cpu(), ref_out, rtol=8e-2, atol=8e-2)
+    torch.testing.assert_close(lse.cpu(), ref_lse, rtol=2e-2, atol=2e-2)
