@@ -31,3 +31,25 @@ for j in range(2):
     av = (out[j] - out[2 + j]).abs().max().item()  # asm vs builtin
     print(f"j={j}: asm-vs-ref bad={ba} maxerr={ma:.4f} | "
           f"builtin-vs-ref bad={bc} maxerr={mc:.4f} | asm-vs-builtin {av:.4f}")
+
+# ---- softmax+pack stream unit ----
+torch.manual_seed(2)
+s_in = (torch.randn(64, 16) * 3).float().cuda()
+ml = torch.stack([torch.randn(64) * 2 - 1, torch.rand(64) * 5 + 0.1], dim=1).cuda()
+# make rows share state across the half-wave (lane l and l+32 carry the same
+# q-row in the real kernel); not required for the unit, it just mirrors use.
+cl2 = 0.125 * 1.44269504
+c_out, ml_out = [t.cpu() for t in ext.probe_gen6_softmax(s_in, ml, cl2)]
+dc = (c_out[0] != c_out[1]).sum().item()
+dml = (ml_out[0] - ml_out[1]).abs().max().item()
+# unpack c as bf16 pairs and compare numerically too (rounding-path slack)
+def unpack(c):
+    lo = (c & 0xffff).to(torch.int32).to(torch.uint16).view(torch.bfloat16).float()
+    hi = (c >> 16).to(torch.int32).to(torch.uint16).view(torch.bfloat16).float()
+    return lo, hi
+a_lo, a_hi = unpack(c_out[0].to(torch.int64))
+b_lo, b_hi = unpack(c_out[1].to(torch.int64))
+dnum = max((a_lo - b_lo).abs().max().item(), (a_hi - b_hi).abs().max().item())
+print(f"GEN6 SOFTMAX: c bit-mismatches={dc} c value-diff={dnum:.6f} "
+      f"m/l/alpha maxdiff={dml:.2e}",
+      "OK" if (dml < 1e-5 and dnum < 1e-2) else "FAIL")
