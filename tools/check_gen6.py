@@ -53,3 +53,12 @@ dnum = max((a_lo - b_lo).abs().max().item(), (a_hi - b_hi).abs().max().item())
 print(f"GEN6 SOFTMAX: c bit-mismatches={dc} c value-diff={dnum:.6f} "
       f"m/l/alpha maxdiff={dml:.2e}",
       "OK" if (dml < 1e-5 and dnum < 1e-2) else "FAIL")
+mm = (c_out[0] != c_out[1])  # (64, 8)
+print("mismatch by c-index:", mm.sum(dim=0).tolist())
+print("mismatch by half-wave: lo", mm[:32].sum().item(), "hi", mm[32:].sum().item())
+l0 = 0
+print("lane0 asm c:", [hex(x) for x in c_out[0][l0].tolist()])
+print("lane0 C   c:", [hex(x) for x in c_out[1][l0].tolist()])
+l0 = 40
+print("lane40 asm c:", [hex(x) for x in c_out[0][l0].tolist()])
+print("lane40 C   c:", [hex(x) for x in c_out[1][l0].tolist()])
