@@ -475,3 +475,15 @@ def test_fp8_narrow_head_padding(ext):
         q.cpu().float(), k.cpu().float(), v.cpu().float(), q_offset=511)
     torch.testing.assert_close(out.cpu(), ref_out, rtol=8e-2, atol=8e-2)
     torch.testing.assert_close(lse.cpu(), ref_lse, rtol=2e-2, atol=2e-2)
+
+
+def test_probe_gen6_softmax_stream(ext):
+    """Generated softmax+pack asm stream: BIT-exact vs the proven C form
+    (fa_prefill5's softmax_pack + pswap) on identical inputs."""
+    torch.manual_seed(2)
+    s_in = (torch.randn(64, 16) * 3).float().cuda()
+    ml = torch.stack([torch.randn(64) * 2 - 1, torch.rand(64) * 5 + 0.1],
+                     dim=1).cuda()
+    c_out, ml_out = ext.probe_gen6_softmax(s_in, ml, 0.125 * 1.44269504)
+    assert (c_out[0] == c_out[1]).all(), "c packs differ"
+    torch.testing.assert_close(ml_out[0], ml_out[1], rtol=0, atol=0)
