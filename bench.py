@@ -84,12 +84,15 @@ def run(rank: int, world: int, args) -> None:
             # at idle clocks and the first timed steps ~15% slow (measured:
             # cold bench 0.349 ms/step vs 0.318 after a prior run on the
             # same box). Spin the SAME step untimed for ~2 s wall clock.
+            # Synchronize every batch of spins so perf_counter tracks DEVICE
+            # time (async launches would otherwise fill the queue and end
+            # the "2 s" ramp after ~0.2 s of real GPU activity — the round-1
+            # cold-box 0.318-0.358 ms/step spread came from that).
             spin_until = time.perf_counter() + 2.0
-            spins = 0
-            while time.perf_counter() < spin_until and spins < 500:
-                step()
-                spins += 1
-            torch.cuda.synchronize()
+            while time.perf_counter() < spin_until:
+                for _ in range(50):
+                    step()
+                torch.cuda.synchronize()
         if args.profile and rank == 0:
             # tracing subsystem: kernel-level chrome trace
             from torch.profiler import ProfilerActivity, profile

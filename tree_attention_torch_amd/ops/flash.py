@@ -43,6 +43,38 @@ def hip_available() -> bool:
     return _load_extension() is not None
 
 
+def decode_loop_chunks(tq: int, g: int, b: int, hq: int) -> int:
+    """Spec-decode routing decision (pure; pinned by tests/test_routing.py).
+
+    Returns the number of decode-kernel query chunks to loop over, or 0 when
+    a single extension dispatch is right (plain decode Tq*G <= 16 — the
+    extension routes that to the split-KV decode kernel — or a prefill-sized
+    batch that fills the 512-block grid).
+
+    Cost model (measured sweep, H=32 @ 32K/128K KV): each decode chunk is
+    one full-bandwidth KV stream; the underfilled prefill launch costs
+    about (512/blocks) streams. Loop while n_chunks is under that.
+    Tq=17..32: 0.72 vs 9.62 ms at 128K; Tq=128: 2.65 vs 9.62; crossover
+    measured at Tq≈256 (prefill 1.47 vs loop 1.61 at 32K) = the n_chunks
+    == 512/blocks point, so strict <.
+    """
+    if g > 16:
+        # MFMA-M batching packs G query heads per KV head into the 16-row
+        # M dimension; G > 16 (e.g. MQA Hq=32, Hkv=1) does not fit one tile.
+        raise ValueError(
+            f"tree_attention: GQA group size Hq/Hkv = {g} exceeds the native "
+            "kernel limit of 16 (MFMA-M batching). Replicate KV heads so "
+            "that Hq/Hkv <= 16."
+        )
+    tq_per = 16 // g
+    n_chunks = -(-tq // tq_per)
+    prefill_blocks = b * hq * (-(-tq // 256))
+    max_chunks = max(4, 512 // max(prefill_blocks, 1))
+    if tq_per < tq and n_chunks < max_chunks and prefill_blocks < 512:
+        return n_chunks
+    return 0
+
+
 def _dispatch_hip(ext, q, k, v, softmax_scale, is_causal, q_offset, kv_offset):
     """Route between the split-KV decode kernel and the prefill kernel.
 
@@ -50,24 +82,13 @@ def _dispatch_hip(ext, q, k, v, softmax_scale, is_causal, q_offset, kv_offset):
     everything else to prefill (256-row q-blocks). For SMALL query batches
     just past 16 rows — speculative decode: Tq 17..64 against a long KV —
     the prefill grid is only B*Hq blocks (vs the 512 the chip wants) with
-    no KV split, so instead loop the decode kernel over <=4 query chunks:
-    each chunk launches the full split-KV grid. Costs n_chunks KV reads,
-    wins whenever the prefill grid would run the chip nearly empty
-    (measured: see profiles/decode_matrix_1gpu.jsonl spec-decode rows).
+    no KV split, so instead loop the decode kernel over query chunks: each
+    chunk launches the full split-KV grid (decode_loop_chunks above).
     """
     tq = q.shape[2]
     g = q.shape[1] // k.shape[1]
-    tq_per = 16 // g
-    n_chunks = -(-tq // tq_per)
-    prefill_blocks = q.shape[0] * q.shape[1] * (-(-tq // 256))
-    # Cost model (measured sweep, H=32 @ 32K/128K KV): each decode chunk is
-    # one full-bandwidth KV stream; the underfilled prefill launch costs
-    # about (512/blocks) streams. Loop while n_chunks is under that.
-    # Tq=17..32: 0.72 vs 9.62 ms at 128K; Tq=128: 2.65 vs 9.62; crossover
-    # measured at Tq≈256 (prefill 1.47 vs loop 1.61 at 32K) = the n_chunks
-    # == 512/blocks point, so strict <.
-    max_chunks = max(4, 512 // max(prefill_blocks, 1))
-    if tq_per < tq and n_chunks < max_chunks and prefill_blocks < 512:
+    if decode_loop_chunks(tq, g, q.shape[0], q.shape[1]) > 0:
+        tq_per = 16 // g
         outs, lses = [], []
         for lo in range(0, tq, tq_per):
             hi = min(tq, lo + tq_per)
@@ -155,4 +176,5 @@ def flash_res_lse(
     return local_attention(q, k, v, softmax_scale=softmax_scale, is_causal=is_causal)
 
 
-__all__ = ["local_attention", "flash_res_lse", "hip_available"]
+__all__ = ["local_attention", "flash_res_lse", "hip_available",
+           "decode_loop_chunks"]

@@ -46,6 +46,13 @@ __all__ = [
 _NEG_CLAMP = -80.0
 
 
+def auto_strategy(out_numel: int) -> str:
+    """Combine-strategy pick (pure; pinned by tests/test_routing.py):
+    allgather when the fp32 payload is latency-bound (< 1 MiB, decode
+    sizes), allreduce when bandwidth-bound (prefill chunks)."""
+    return "allgather" if out_numel * 4 < (1 << 20) else "allreduce"
+
+
 def _rescale(out: torch.Tensor, lse: torch.Tensor, m: torch.Tensor):
     """num = out * w, den = w with w = exp(lse - m), safe for lse = m = -inf."""
     w = torch.exp(torch.clamp(lse - m, min=_NEG_CLAMP, max=0.0))
@@ -206,8 +213,7 @@ def tree_combine(
             return _Handle()
         return out.float(), lse.float()
     if strategy == "auto":
-        payload = out.numel() * 4
-        strategy = "allgather" if payload < (1 << 20) else "allreduce"
+        strategy = auto_strategy(out.numel())
     if strategy == "allgather":
         return tree_combine_allgather(out, lse, group=group, async_op=async_op)
     if strategy == "allreduce":

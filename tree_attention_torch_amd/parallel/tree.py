@@ -27,6 +27,21 @@ from .combine import tree_combine
 __all__ = ["tree_decode", "tree_attention", "TreeAttention"]
 
 
+def default_q_chunk(tq: int, b: int, hq: int) -> int:
+    """Prefill chunk size (pure; pinned by tests/test_routing.py).
+
+    Keep every chunk's grid full: the prefill kernel launches
+    B*Hq*(chunk/256) blocks and the chip wants >= 512 (a 4096-row chunk at
+    B*Hq=8 is 128 blocks — measured 5.5x slower than unchunked). When
+    chunks DO fill the grid, chunking is mildly faster even at world 1
+    (causal-tail scheduling: 13.4 vs 14.2 ms at H=32, 32K rows, same box)
+    and at world > 1 it is what overlaps each chunk's collective with the
+    next kernel.
+    """
+    min_chunk = max(4096, (512 * 256) // max(b * hq, 1))
+    return tq if tq <= min_chunk else min_chunk
+
+
 def _rank_and_world(group) -> tuple[int, int]:
     if dist.is_available() and dist.is_initialized():
         return dist.get_rank(group), dist.get_world_size(group)
@@ -90,16 +105,7 @@ def tree_attention(
         softmax_scale = 1.0 / math.sqrt(q.shape[-1])
 
     if q_chunk is None:
-        # Keep every chunk's grid full: the prefill kernel launches
-        # B*Hq*(chunk/256) blocks and the chip wants >= 512 (a 4096-row
-        # chunk at B*Hq=8 is 128 blocks — measured 5.5x slower than
-        # unchunked). When chunks DO fill the grid, chunking is mildly
-        # faster even at world 1 (causal-tail scheduling: 13.4 vs
-        # 14.2 ms at H=32, 32K rows, same box) and at world > 1 it is
-        # what overlaps each chunk's collective with the next kernel.
-        b, hq = q.shape[0], q.shape[1]
-        min_chunk = max(4096, (512 * 256) // max(b * hq, 1))
-        q_chunk = tq if tq <= min_chunk else min_chunk
+        q_chunk = default_q_chunk(tq, q.shape[0], q.shape[1])
 
     if tq <= q_chunk:
         out_l, lse_l = local_attention(

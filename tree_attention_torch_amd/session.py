@@ -93,9 +93,23 @@ class DecodeSession:
         self.total += 1
 
     def _local_partial(self, q: torch.Tensor, softmax_scale: float | None):
-        if q.device.type == "cuda" and self.k.size(3) == 128:
-            # the zero-copy cache binding is head_dim-128 native; other dims
-            # take the generic path below (padding inside local_attention)
+        # zero-copy eligibility mirrors the binding's TORCH_CHECKs
+        # (bindings.cpp flash_attention_cache): D=128; q bf16/fp16; cache
+        # dtype == q dtype or fp8 cache under a bf16 q; (Hq/Hkv)*Tq <= 16
+        # (MFMA-M batching). Anything else falls through to the generic
+        # slice + local_attention path, which handles those cases.
+        zero_copy_ok = (
+            q.device.type == "cuda"
+            and self.k.size(3) == 128
+            and q.dtype in (torch.bfloat16, torch.float16)
+            and (
+                self.k.dtype == q.dtype
+                or (self.k.dtype == torch.float8_e4m3fn
+                    and q.dtype == torch.bfloat16)
+            )
+            and (q.shape[1] // self.k.shape[1]) * q.shape[2] <= 16
+        )
+        if zero_copy_ok:
             # zero-copy cache path: the kernel takes the cache's head stride
             # and reads the LIVE length from a device scalar, so this call
             # is hipGraph-capturable (graphed_attend) and never copies KV.
@@ -144,7 +158,7 @@ class DecodeSession:
         """Capture the decode-attend into a hipGraph (world_size 1).
 
         Returns (replay, out): write the query into q_static, keep
-        self._len_dev fresh via attend_sync_len(), call replay(), read out.
+        self._len_dev fresh via sync_len(), call replay(), read out.
         The kernel reads the live KV length from device memory, so ONE
         captured graph serves the whole growing sequence — no recapture.
         """
