@@ -154,15 +154,23 @@ class DecodeSession:
         return out
 
     def graphed_attend(self, q_static: torch.Tensor,
-                       softmax_scale: float | None = None):
-        """Capture the decode-attend into a hipGraph (world_size 1).
+                       softmax_scale: float | None = None,
+                       combine: str = "auto"):
+        """Capture the decode-attend into a hipGraph.
 
         Returns (replay, out): write the query into q_static, keep
-        self._len_dev fresh via sync_len(), call replay(), read out.
+        self._len_dev fresh via sync_len(), call replay() -> output tensor.
         The kernel reads the live KV length from device memory, so ONE
         captured graph serves the whole growing sequence — no recapture.
+
+        world_size 1: the whole step replays from the graph and ``out`` is
+        the static output buffer. world_size > 1: the LOCAL partial replays
+        from the graph (launch-bound part of the step) and the cross-rank
+        combine collective runs eagerly per call — RCCL collectives are not
+        capturable into a per-rank local graph, but they are a single
+        latency-bound call on a ~16 KB payload; ``out`` is None and
+        replay()'s return value is the fresh combined output.
         """
-        assert self.world == 1, "graphed_attend covers the local step"
         assert self.device.type == "cuda"
         # warmup on a side stream (allocator + kernels), then capture
         s = torch.cuda.Stream()
@@ -173,14 +181,23 @@ class DecodeSession:
         torch.cuda.current_stream().wait_stream(s)
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
-            out_l, _ = self._local_partial(q_static, softmax_scale)
+            out_l, lse_l = self._local_partial(q_static, softmax_scale)
         self._graph = graph  # keep alive
 
-        def replay() -> torch.Tensor:
-            graph.replay()
-            return out_l
+        if self.world == 1:
+            def replay() -> torch.Tensor:
+                graph.replay()
+                return out_l
 
-        return replay, out_l
+            return replay, out_l
+
+        def replay_dist() -> torch.Tensor:
+            graph.replay()
+            out, _ = tree_combine(out_l, lse_l, strategy=combine,
+                                  group=self.group)
+            return out
+
+        return replay_dist, None
 
     def sync_len(self) -> None:
         """Refresh the device-side length after append()s (graphed path)."""
