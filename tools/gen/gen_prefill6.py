@@ -1,16 +1,37 @@
 #!/usr/bin/env python3
-"""Generate the hand-placed asm interior phase for fa_prefill6 (gfx950).
+"""Generate the hand-placed asm streams for fa_prefill6 (gfx950).
 
 Emits tree_attention_torch_amd/ops/hip/fa_prefill6_gen.h, included by
-fa_kernels.hip between GEN6 markers. The stream follows the guide's
-pwg4x64 stagger: per 32-key block nb, startSM(nb) fills PV(nb-1)'s MFMA
-gaps and finishSM(nb) fills QK^T(nb+1)'s; all DS ops are in-asm with
-counted lgkmcnt waits. Geometry = fa_prefill4: 4 waves (1/SIMD), two
-32-row q-blocks per wave, KVBLK=128 staged by glds (C code, vmcnt-counted
-outside the asm).
+fa_kernels.hip. The prefill6 kernel is the 1-wave/SIMD hand-scheduled
+regime (the guide's pwg4x64 direction): 4 waves per 256-row q-block, each
+wave two 32-row q-blocks, whole 512-register file per lane. The per-TILE
+main loop body is ONE asm volatile statement so no hand-owned register
+state ever crosses a statement boundary the compiler could interleave
+with; O accumulators live in literal AGPRs a[0:127] (clobbered), Q
+fragments are operand-passed AGPRs, scores/packs/temps are clobbered
+fixed arch VGPRs v[144:255], and m/l state crosses tiles as "+v"
+operands. C code around the statements does staging (glds, counted
+vmcnt), addressing, the boundary-tile fallback and the epilogue.
 
-Stage 1 (this file's current state): QKT_PAIR unit — the 16-MFMA
-double-chain with in-asm pipelined ds_read_b128 K reads.
+Verified silicon rules baked in (docs/ROUND1_NOTES.md):
+  * MFMA D -> VALU reader: 12 wait states (s_nop 11 at stream seams);
+    D-as-C accumulate chains need none; never pad inside a chain.
+  * VGPR write -> v_permlane* reader: s_nop 1.
+  * v_exp_f32 (TRANS) result -> reader: one wait state.
+  * ds_read in-asm with counted lgkmcnt (one-ahead pipelining).
+  * The XOR swizzles are nb-independent: K/V slice addresses are
+    lane-resolved registers + an immediate ds offset per (nb, ks).
+  * Multi-instruction outputs need "=&v" earlyclobber.
+
+Register map (fixed, all clobbered by the tile statement):
+  a[0:127]    O accumulators: a[j*64 + nd*16 + r]
+  v[192:207]  sjA0   v[208:223] sjA1   (scores, buffer A)
+  v[224:239]  sjB0   v[240:255] sjB1   (scores, buffer B — stagger)
+  v[176:183]  c0 packs   v[184:191] c1 packs
+  v[160:163]  kr0        v[164:167] kr1   (K read pipeline)
+  v[144:155]  vr: three 4-reg V-fragment slots (tr-read pipeline)
+  v[168:175]  temps: T0 T1 MT0 MT1 RS P0 P1 spare
+  v[156:159]  AL0 AL1 spare spare
 """
 
 import os
@@ -21,92 +42,425 @@ OUT = os.path.join(os.path.dirname(__file__), "..", "..",
 
 MFMA = "v_mfma_f32_32x32x16_bf16"
 
+# ---- fixed register map ----
+SJ = {("A", 0): 192, ("A", 1): 208, ("B", 0): 224, ("B", 1): 240}
+CP = {0: 176, 1: 184}
+KR = {0: 160, 1: 164}
+VRSLOT = [144, 148, 152]          # three 4-reg V fragment slots
+T0, T1 = "v168", "v169"
+MT = {0: "v170", 1: "v171"}
+RS, P0, P1 = "v172", "v173", "v174"
+AL = {0: "v156", 1: "v157"}
 
-def qkt_pair_stream():
-    """16 MFMAs (two chains j0/j1 sharing K), K via pipelined in-asm
-    ds_read_b128: one read in flight ahead, lgkmcnt(1) waits.
+# tile-statement operand indices
+OP_M = {0: "%0", 1: "%1"}
+OP_L = {0: "%2", 1: "%3"}
+OP_KA = lambda s: f"%{4 + s}"          # noqa: E731
+OP_VA = lambda c, nd: f"%{12 + c * 4 + nd}"   # noqa: E731
+OP_CL2 = "%20"
+OP_Q = lambda j, s: f"%{21 + j * 8 + s}"      # noqa: E731
 
-    asm operands:
-      %0  = sj0 (+v f32x16)   %1 = sj1 (+v f32x16)
-      %2  = kr0 (+v V8 scratch) %3 = kr1 (+v V8 scratch)
-      %4..%11  = k slice addresses (v, bytes, lane-resolved, slice 0..7)
-      %12..%19 = q_frag j0 slices 0..7 (a, V8)
-      %20..%27 = q_frag j1 slices 0..7 (a, V8)
-    The nb offset is an immediate: offset:OFF (= nb*8192).
-    """
+
+def vr(base, n=4):
+    return f"v[{base}:{base + n - 1}]"
+
+
+def sj_reg(buf, j, i):
+    return f"v{SJ[(buf, j)] + i}"
+
+
+# ---------------------------------------------------------------------------
+# Instruction-stream builders. Each returns a list of (text, kind) where
+# kind in {"mfma", "valu", "trans", "ds", "wait", "nop", "salu"} — the v2
+# scheduler interleaves by kind; v1 just concatenates.
+# ---------------------------------------------------------------------------
+
+def qkt_stream(nb, buf):
+    """QK^T for both q-blocks over 32-key block nb: 16 MFMAs, K slices via
+    in-asm pipelined ds_read_b128 (one read ahead, counted lgkmcnt).
+    Writes sj{buf}0 / sj{buf}1 (fixed v-ranges); A = kr pipeline; B = Q
+    operand AGPRs. Probe-verified stream (probe_gen6_qkt)."""
+    off = nb * 8192
+    s0 = f"v[{SJ[(buf, 0)]}:{SJ[(buf, 0)] + 15}]"
+    s1 = f"v[{SJ[(buf, 1)]}:{SJ[(buf, 1)] + 15}]"
+    kr0, kr1 = vr(KR[0]), vr(KR[1])
     L = []
-    # prologue: two reads in flight
+    L.append((f"ds_read_b128 {kr0}, {OP_KA(0)} offset:{off}", "ds"))
+    L.append((f"ds_read_b128 {kr1}, {OP_KA(1)} offset:{off}", "ds"))
+    L.append(("s_waitcnt lgkmcnt(1)", "wait"))
+    L.append((f"{MFMA} {s0}, {kr0}, {OP_Q(0, 0)}, 0", "mfma"))
+    L.append((f"{MFMA} {s1}, {kr0}, {OP_Q(1, 0)}, 0", "mfma"))
+    for s in range(1, 8):
+        cur = kr1 if (s % 2) else kr0
+        nxt = kr0 if (s % 2) else kr1
+        if s + 1 < 8:
+            L.append((f"ds_read_b128 {nxt}, {OP_KA(s + 1)} offset:{off}", "ds"))
+            L.append(("s_waitcnt lgkmcnt(1)", "wait"))
+        else:
+            L.append(("s_waitcnt lgkmcnt(0)", "wait"))
+        L.append((f"{MFMA} {s0}, {cur}, {OP_Q(0, s)}, {s0}", "mfma"))
+        L.append((f"{MFMA} {s1}, {cur}, {OP_Q(1, s)}, {s1}", "mfma"))
+    return L
+
+
+def sm_start_stream(j, buf):
+    """startSM: in-lane + cross-half max of sj{buf}{j}, m_new (exp2 domain),
+    alpha; updates m_run. MT{j} carries m_new to finishSM. Stream is the
+    probe-verified SOFTMAX_PACK head with physical registers."""
+    s = lambda i: sj_reg(buf, j, i)  # noqa: E731
+    mt, al, m = MT[j], AL[j], OP_M[j]
+    L = [
+        (f"v_max3_f32 {mt}, {s(0)}, {s(1)}, {s(2)}", "valu"),
+        (f"v_max3_f32 {T0}, {s(3)}, {s(4)}, {s(5)}", "valu"),
+        (f"v_max3_f32 {T1}, {s(6)}, {s(7)}, {s(8)}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {T0}, {T1}", "valu"),
+        (f"v_max3_f32 {T0}, {s(9)}, {s(10)}, {s(11)}", "valu"),
+        (f"v_max3_f32 {T1}, {s(12)}, {s(13)}, {s(14)}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {T0}, {T1}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {s(15)}, {s(15)}", "valu"),
+        (f"v_mov_b32 {T0}, {mt}", "valu"),
+        (f"v_mov_b32 {T1}, {mt}", "valu"),
+        ("s_nop 1", "nop"),  # VGPR write -> v_permlane hazard
+        (f"v_permlane32_swap_b32 {T0}, {T1}", "valu"),
+        (f"v_max_f32 {mt}, {T0}, {T1}", "valu"),
+        (f"v_mul_f32 {mt}, {mt}, {OP_CL2}", "valu"),
+        (f"v_max_f32 {mt}, {m}, {mt}", "valu"),          # m_new
+        (f"v_sub_f32 {T0}, {m}, {mt}", "valu"),
+        (f"v_exp_f32 {al}, {T0}", "trans"),              # alpha
+        (f"v_mov_b32 {m}, {mt}", "valu"),                # m_run = m_new
+    ]
+    return L
+
+
+def sm_finish_stream(j, buf):
+    """finishSM: exp2+pack of sj{buf}{j} against MT{j}, row sum, l update,
+    pswap of the c{j} packs. Probe-verified tail with physical registers."""
+    s = lambda i: sj_reg(buf, j, i)  # noqa: E731
+    mt, al, l = MT[j], AL[j], OP_L[j]
+    c = lambda i: f"v{CP[j] + i}"  # noqa: E731
+    L = [(f"v_mov_b32 {RS}, 0", "valu")]
+    for i in range(8):
+        L += [
+            (f"v_fma_f32 {P0}, {s(2 * i)}, {OP_CL2}, -{mt}", "valu"),
+            (f"v_fma_f32 {P1}, {s(2 * i + 1)}, {OP_CL2}, -{mt}", "valu"),
+            (f"v_exp_f32 {P0}, {P0}", "trans"),
+            (f"v_exp_f32 {P1}, {P1}", "trans"),
+            ("s_nop 0", "nop"),  # TRANS result needs one state
+            (f"v_cvt_pk_bf16_f32 {c(i)}, {P0}, {P1}", "valu"),
+            (f"v_add_f32 {P0}, {P0}, {P1}", "valu"),
+            (f"v_add_f32 {RS}, {RS}, {P0}", "valu"),
+        ]
+    L += [
+        (f"v_mov_b32 {T0}, {RS}", "valu"),
+        (f"v_mov_b32 {T1}, {RS}", "valu"),
+        ("s_nop 1", "nop"),
+        (f"v_permlane32_swap_b32 {T0}, {T1}", "valu"),
+        (f"v_add_f32 {RS}, {T0}, {T1}", "valu"),
+        (f"v_fma_f32 {l}, {l}, {al}, {RS}", "valu"),      # l = l*alpha + rs
+        ("s_nop 1", "nop"),
+        (f"v_permlane32_swap_b32 {c(0)}, {c(2)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(1)}, {c(3)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(4)}, {c(6)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(5)}, {c(7)}", "valu"),
+    ]
+    return L
+
+
+def rescale_call(j, label):
+    """Vote-skipped O rescale call site: if any lane's alpha != 1, call the
+    per-statement shared rescale subroutine for block j (the 200-instr
+    body would blow L1i if inlined at every nb; the vote fires rarely —
+    the running max stops moving after early tiles)."""
+    return [
+        (f"v_cmp_neq_f32 vcc, 1.0, {AL[j]}", "valu"),
+        ("s_nop 0", "nop"),  # VALU-writes-VCC -> branch-on-VCCZ guard
+        (f"s_cbranch_vccz TA_G6_RSD{label}_{j}", "salu"),
+        (f"s_call_b64 s[40:41], TA_G6_RSFN_{j}", "salu"),
+        (f"TA_G6_RSD{label}_{j}:", "label"),
+    ]
+
+
+def rescale_fn(j):
+    """The shared rescale body: O[j] *= alpha_j via v_accvgpr round trip
+    (gfx950 VALU cannot take AGPR operands directly; measured compile
+    error). Emitted once per tile statement, jumped over in fall-through."""
+    al = AL[j]
+    L = [(f"TA_G6_RSFN_{j}:", "label"),
+         ("s_nop 11", "nop")]  # PV MFMA D -> accvgpr_read (rare path)
+    base = j * 64
+    for r in range(0, 64, 4):
+        for u in range(4):
+            L.append((f"v_accvgpr_read_b32 v{168 + u}, a{base + r + u}", "valu"))
+        for u in range(4):
+            L.append((f"v_mul_f32 v{168 + u}, v{168 + u}, {al}", "valu"))
+        for u in range(4):
+            L.append((f"v_accvgpr_write_b32 a{base + r + u}, v{168 + u}", "valu"))
+    L.append(("s_nop 1", "nop"))  # accvgpr_write -> MFMA operand guard
+    L.append(("s_setpc_b64 s[40:41]", "salu"))
+    return L
+
+
+def rescale_tail():
+    """Both rescale subroutines + the jump over them (statement tail)."""
+    L = [("s_branch TA_G6_RSEND", "salu")]
+    L += rescale_fn(0)
+    L += rescale_fn(1)
+    L.append(("TA_G6_RSEND:", "label"))
+    return L
+
+
+def pv_stream(nb):
+    """PV for 32-key block nb: 16 tr-reads (pipelined, one fragment ahead)
+    + 16 MFMAs accumulating into a[]. A = V fragment (two adjacent
+    tr-reads = bf16x8), B = c packs (pswap'd), D/C literal AGPRs."""
+    L = []
+    frags = [(ks, nd) for ks in range(2) for nd in range(4)]
+
+    def emit_reads(fi):
+        ks, nd = frags[fi]
+        slot = VRSLOT[fi % 3]
+        off = nb * 8192 + ks * 4096
+        L.append((f"ds_read_b64_tr_b16 {vr(slot, 2)}, {OP_VA(0, nd)} offset:{off}", "ds"))
+        L.append((f"ds_read_b64_tr_b16 {vr(slot + 2, 2)}, {OP_VA(1, nd)} offset:{off}", "ds"))
+
+    emit_reads(0)
+    emit_reads(1)
+    for fi in range(8):
+        ks, nd = frags[fi]
+        slot = VRSLOT[fi % 3]
+        # before MFMA(fi): outstanding = f(fi) + f(fi+1) (4 reads); waiting
+        # to <=2 leaves f(fi+1) in flight with f(fi) landed
+        if fi < 7:
+            L.append(("s_waitcnt lgkmcnt(2)", "wait"))
+        else:
+            L.append(("s_waitcnt lgkmcnt(0)", "wait"))
+        a = vr(slot)
+        L.append((f"{MFMA} a[{nd * 16}:{nd * 16 + 15}], {a}, "
+                  f"v[{CP[0] + ks * 4}:{CP[0] + ks * 4 + 3}], "
+                  f"a[{nd * 16}:{nd * 16 + 15}]", "mfma"))
+        L.append((f"{MFMA} a[{64 + nd * 16}:{64 + nd * 16 + 15}], {a}, "
+                  f"v[{CP[1] + ks * 4}:{CP[1] + ks * 4 + 3}], "
+                  f"a[{64 + nd * 16}:{64 + nd * 16 + 15}]", "mfma"))
+        if fi + 2 < 8:
+            emit_reads(fi + 2)
+    return L
+
+
+# ---------------------------------------------------------------------------
+# v1 tile body: sequential phases per nb (correctness-first)
+# ---------------------------------------------------------------------------
+
+def tile_body_v1():
+    L = []
+    for nb in range(4):
+        L += qkt_stream(nb, "A")
+        L.append(("s_nop 11", "nop"))   # QKT D -> softmax VALU reader
+        L += sm_start_stream(0, "A")
+        L += sm_finish_stream(0, "A")
+        L += sm_start_stream(1, "A")
+        L += sm_finish_stream(1, "A")
+        L += rescale_call(0, f"V1N{nb}")
+        L += rescale_call(1, f"V1N{nb}")
+        L += pv_stream(nb)
+        if nb < 3:
+            L.append(("s_nop 11", "nop"))  # PV D -> (rare) rescale reader
+    L += rescale_tail()
+    return L
+
+
+# ---------------------------------------------------------------------------
+# v2 tile body: the stagger schedule —
+#   ramp:   QKT(0) | QKT(1)+SM(0)
+#   steady: PV(nb-1)||startSM(nb) -> QKT(nb+1)||finishSM(nb)
+#   drain:  finishSM(3) exposed, PV(3)
+# ---------------------------------------------------------------------------
+
+def interleave(mfma_stream, fills, max_per_gap=5):
+    """Place `fills` into the issue gaps of `mfma_stream`: after each MFMA,
+    up to max_per_gap fill items (preserving each stream's own order).
+    ds/wait items of the MFMA stream keep their position relative to their
+    MFMAs. Left-over fills are appended after the stream."""
+    out = []
+    fi = 0
+    for item in mfma_stream:
+        out.append(item)
+        if item[1] == "mfma":
+            placed = 0
+            while fi < len(fills) and placed < max_per_gap:
+                out.append(fills[fi])
+                fi += 1
+                if fills[fi - 1][1] != "nop":
+                    placed += 1
+    out.extend(fills[fi:])
+    return out
+
+
+def tile_body_v2():
+    L = []
+    # ramp: QKT(0) -> sjA; QKT(1) -> sjB with SM(0) as fills
+    L += qkt_stream(0, "A")
+    L.append(("s_nop 11", "nop"))
+    sm0 = sm_start_stream(0, "A") + sm_finish_stream(0, "A") + \
+        sm_start_stream(1, "A") + sm_finish_stream(1, "A")
+    L += interleave(qkt_stream(1, "B"), sm0)
+    L += rescale_call(0, "V2R") + rescale_call(1, "V2R")
+    # steady state over nb = 1..3
+    for nb in range(1, 4):
+        cur = "B" if (nb % 2) else "A"    # sj buffer holding scores(nb)
+        nxt = "A" if (nb % 2) else "B"
+        # phase X: PV(nb-1) || startSM(nb) both blocks
+        startf = sm_start_stream(0, cur) + sm_start_stream(1, cur)
+        L += interleave(pv_stream(nb - 1), startf)
+        L.append(("s_nop 11", "nop"))
+        # phase Y: QKT(nb+1) || finishSM(nb); last nb has no QKT(4)
+        finishf = sm_finish_stream(0, cur) + sm_finish_stream(1, cur)
+        if nb < 3:
+            L += interleave(qkt_stream(nb + 1, nxt), finishf)
+        else:
+            L += finishf
+        L += rescale_call(0, f"V2N{nb}") + rescale_call(1, f"V2N{nb}")
+    L.append(("s_nop 1", "nop"))
+    L += pv_stream(3)
+    L += rescale_tail()
+    return L
+
+
+# ---------------------------------------------------------------------------
+# emission
+# ---------------------------------------------------------------------------
+
+def render(body_items):
+    lines = []
+    for text, kind in body_items:
+        if kind == "label":
+            lines.append(text)      # labels are not tab-indented
+        else:
+            lines.append(text)
+    return "\\n\\t".join(lines)
+
+
+def clobbers():
+    regs = [f'"a{i}"' for i in range(128)]
+    regs += [f'"v{i}"' for i in range(144, 256)]
+    regs += ['"s40"', '"s41"']  # s_call_b64 return address
+    regs += ['"vcc"', '"memory"']
+    return ", ".join(regs)
+
+
+def tile_macro(name, body):
+    """One asm statement: counted vmcnt wait (the caller pre-drains with a
+    separate vmcnt(0) statement when no next-tile stage is in flight),
+    barrier, 4 nb phases, shared rescale subroutines, tail barrier.
+    Labels get a per-expansion suffix via %=."""
+    body = body.replace("TA_G6_RS", "TA_G6_%=_RS")
+    return f'''#define {name}(m0, m1, l0, l1, ka, va, cl2, qa, qb)                 \\
+  asm volatile(                                                             \\
+      "s_waitcnt vmcnt(16)\\n\\t"                                            \\
+      "s_barrier\\n\\t"                                                      \\
+      "{body}\\n\\t"                                                         \\
+      "s_barrier"                                                           \\
+      : "+v"(m0), "+v"(m1), "+v"(l0), "+v"(l1)                              \\
+      : "v"(ka[0]), "v"(ka[1]), "v"(ka[2]), "v"(ka[3]), "v"(ka[4]),         \\
+        "v"(ka[5]), "v"(ka[6]), "v"(ka[7]),                                 \\
+        "v"(va[0]), "v"(va[1]), "v"(va[2]), "v"(va[3]), "v"(va[4]),         \\
+        "v"(va[5]), "v"(va[6]), "v"(va[7]),                                 \\
+        "v"(cl2),                                                           \\
+        "a"(qa[0]), "a"(qa[1]), "a"(qa[2]), "a"(qa[3]), "a"(qa[4]),         \\
+        "a"(qa[5]), "a"(qa[6]), "a"(qa[7]),                                 \\
+        "a"(qb[0]), "a"(qb[1]), "a"(qb[2]), "a"(qb[3]), "a"(qb[4]),         \\
+        "a"(qb[5]), "a"(qb[6]), "a"(qb[7])                                  \\
+      : {clobbers()})
+'''
+
+
+def zero_o_macro():
+    writes = "\\n\\t".join(
+        f"v_accvgpr_write_b32 a{i}, 0" for i in range(128))
+    return f'''#define TA_GEN6_ZERO_O()                                          \\
+  asm volatile("{writes}" ::: {", ".join(f'"a{i}"' for i in range(128))})
+'''
+
+
+def read_o_macros():
+    out = []
+    for j in range(2):
+        for nd in range(4):
+            base = j * 64 + nd * 16
+            reads = "\\n\\t".join(
+                f"v_accvgpr_read_b32 %{i}, a{base + i}" for i in range(16))
+            outs = ", ".join(f'"=v"(f[{i}])' for i in range(16))
+            out.append(
+                f'''#define TA_GEN6_READ_O_J{j}_ND{nd}(f)                               \\
+  asm volatile("s_nop 11\\n\\t{reads}" : {outs})
+''')
+    return "\n".join(out)
+
+
+# ---------------------------------------------------------------------------
+# probe macros (unit probes in fa_kernels.hip reference these — keep the
+# original operand-based forms verified in round 1)
+# ---------------------------------------------------------------------------
+
+def probe_qkt_pair_stream():
+    L = []
     L.append("ds_read_b128 %2, %4 offset:OFF")
     L.append("ds_read_b128 %3, %5 offset:OFF")
-    L.append("s_waitcnt lgkmcnt(1)")           # slice 0 landed
+    L.append("s_waitcnt lgkmcnt(1)")
     L.append(f"{MFMA} %0, %2, %12, 0")
     L.append(f"{MFMA} %1, %2, %20, 0")
     for s in range(1, 8):
-        kr_cur = 3 if (s % 2) else 2           # reg holding slice s
+        kr_cur = 3 if (s % 2) else 2
         kr_nxt = 2 if (s % 2) else 3
         if s + 1 < 8:
             L.append(f"ds_read_b128 %{kr_nxt}, %{4 + s + 1} offset:OFF")
-            L.append("s_waitcnt lgkmcnt(1)")   # slice s landed
+            L.append("s_waitcnt lgkmcnt(1)")
         else:
             L.append("s_waitcnt lgkmcnt(0)")
         L.append(f"{MFMA} %0, %{kr_cur}, %{12 + s}, %0")
         L.append(f"{MFMA} %1, %{kr_cur}, %{20 + s}, %1")
-    L.append("s_nop 11")                        # last MFMA D -> VALU reader
+    L.append("s_nop 11")
     return L
 
 
-def softmax_pack_stream():
-    """startSM+finishSM for one 32-key block, exp2 domain, matching
-    fa_prefill5's softmax_pack + pswap exactly.
-
-    Operands: %0..%7 c-packs (=&v u32); %8 alpha (=&v); %9..%14 temps
-    t0,t1,mt,rs,p0,p1 (=&v); %15 m_run (+v); %16 l_run (+v);
-    inputs %17..%32 = s0..s15 (v), %33 = cl2 (v).
-    Cross-half exchange via the permlane32_swap dup trick:
-    o0=x; o1=x; swap(o0,o1) -> o0 = lo-half dup, o1 = hi-half dup.
-    """
+def probe_softmax_pack_stream():
     L = []
-    T0, T1, MT, RS, P0, P1, M, LR, CL = (
+    T0p, T1p, MTp, RSp, P0p, P1p, M, LR, CL = (
         "%9", "%10", "%11", "%12", "%13", "%14", "%15", "%16", "%33")
-    sreg = lambda i: f"%{17 + i}"
-    # in-lane max tree (16 values): 7 max3 + dup for s15
-    L += [f"v_max3_f32 {MT}, {sreg(0)}, {sreg(1)}, {sreg(2)}",
-          f"v_max3_f32 {T0}, {sreg(3)}, {sreg(4)}, {sreg(5)}",
-          f"v_max3_f32 {T1}, {sreg(6)}, {sreg(7)}, {sreg(8)}",
-          f"v_max3_f32 {MT}, {MT}, {T0}, {T1}",
-          f"v_max3_f32 {T0}, {sreg(9)}, {sreg(10)}, {sreg(11)}",
-          f"v_max3_f32 {T1}, {sreg(12)}, {sreg(13)}, {sreg(14)}",
-          f"v_max3_f32 {MT}, {MT}, {T0}, {T1}",
-          f"v_max3_f32 {MT}, {MT}, {sreg(15)}, {sreg(15)}"]
-    # cross-half max
-    L += [f"v_mov_b32 {T0}, {MT}",
-          f"v_mov_b32 {T1}, {MT}",
-          "s_nop 1",  # VGPR write -> v_permlane* hazard (guide T21 rule)
-          f"v_permlane32_swap_b32 {T0}, {T1}",
-          f"v_max_f32 {MT}, {T0}, {T1}",
-          f"v_mul_f32 {MT}, {MT}, {CL}",
-          f"v_max_f32 {MT}, {M}, {MT}",          # m_new
-          f"v_sub_f32 {T0}, {M}, {MT}",
-          f"v_exp_f32 %8, {T0}",                  # alpha (read much later)
-          f"v_mov_b32 {M}, {MT}",                 # m_run = m_new
-          f"v_mov_b32 {RS}, 0"]
-    for i in range(8):
-        L += [f"v_fma_f32 {P0}, {sreg(2 * i)}, {CL}, -{MT}",
-              f"v_fma_f32 {P1}, {sreg(2 * i + 1)}, {CL}, -{MT}",
-              f"v_exp_f32 {P0}, {P0}",
-              f"v_exp_f32 {P1}, {P1}",
-              "s_nop 0",  # TRANS result latency: v_exp -> reader needs 1 state
-              f"v_cvt_pk_bf16_f32 %{i}, {P0}, {P1}",
-              f"v_add_f32 {P0}, {P0}, {P1}",
-              f"v_add_f32 {RS}, {RS}, {P0}"]
-    # cross-half rowsum
-    L += [f"v_mov_b32 {T0}, {RS}",
-          f"v_mov_b32 {T1}, {RS}",
+    sreg = lambda i: f"%{17 + i}"  # noqa: E731
+    L += [f"v_max3_f32 {MTp}, {sreg(0)}, {sreg(1)}, {sreg(2)}",
+          f"v_max3_f32 {T0p}, {sreg(3)}, {sreg(4)}, {sreg(5)}",
+          f"v_max3_f32 {T1p}, {sreg(6)}, {sreg(7)}, {sreg(8)}",
+          f"v_max3_f32 {MTp}, {MTp}, {T0p}, {T1p}",
+          f"v_max3_f32 {T0p}, {sreg(9)}, {sreg(10)}, {sreg(11)}",
+          f"v_max3_f32 {T1p}, {sreg(12)}, {sreg(13)}, {sreg(14)}",
+          f"v_max3_f32 {MTp}, {MTp}, {T0p}, {T1p}",
+          f"v_max3_f32 {MTp}, {MTp}, {sreg(15)}, {sreg(15)}"]
+    L += [f"v_mov_b32 {T0p}, {MTp}",
+          f"v_mov_b32 {T1p}, {MTp}",
           "s_nop 1",
-          f"v_permlane32_swap_b32 {T0}, {T1}",
-          f"v_add_f32 {RS}, {T0}, {T1}",
-          f"v_fma_f32 {LR}, {LR}, %8, {RS}"]      # l = l*alpha + rs
-    # pswap: pairs (0,2) (1,3) (4,6) (5,7)
+          f"v_permlane32_swap_b32 {T0p}, {T1p}",
+          f"v_max_f32 {MTp}, {T0p}, {T1p}",
+          f"v_mul_f32 {MTp}, {MTp}, {CL}",
+          f"v_max_f32 {MTp}, {M}, {MTp}",
+          f"v_sub_f32 {T0p}, {M}, {MTp}",
+          f"v_exp_f32 %8, {T0p}",
+          f"v_mov_b32 {M}, {MTp}",
+          f"v_mov_b32 {RSp}, 0"]
+    for i in range(8):
+        L += [f"v_fma_f32 {P0p}, {sreg(2 * i)}, {CL}, -{MTp}",
+              f"v_fma_f32 {P1p}, {sreg(2 * i + 1)}, {CL}, -{MTp}",
+              f"v_exp_f32 {P0p}, {P0p}",
+              f"v_exp_f32 {P1p}, {P1p}",
+              "s_nop 0",
+              f"v_cvt_pk_bf16_f32 %{i}, {P0p}, {P1p}",
+              f"v_add_f32 {P0p}, {P0p}, {P1p}",
+              f"v_add_f32 {RSp}, {RSp}, {P0p}"]
+    L += [f"v_mov_b32 {T0p}, {RSp}",
+          f"v_mov_b32 {T1p}, {RSp}",
+          "s_nop 1",
+          f"v_permlane32_swap_b32 {T0p}, {T1p}",
+          f"v_add_f32 {RSp}, {T0p}, {T1p}",
+          f"v_fma_f32 {LR}, {LR}, %8, {RSp}"]
     L += ["s_nop 1",
           "v_permlane32_swap_b32 %0, %2",
           "v_permlane32_swap_b32 %1, %3",
@@ -115,30 +469,8 @@ def softmax_pack_stream():
     return L
 
 
-def emit():
-    qkt = "\\n\\t".join(qkt_pair_stream())
-    src = f'''// GENERATED by tools/gen/gen_prefill6.py — do not edit by hand.
-#pragma once
-
-// QKT_PAIR(nb): sj0/sj1 = K_tile(nb)^T-swapped MFMA chains for both
-// q-blocks, K read in-asm (pipelined ds_read_b128, one ahead).
-// ka0..ka7: per-lane byte address of slice s at nb=0 (the XOR swizzle is
-// nb-independent because nb*32 is 0 mod 16); nb advances by an immediate
-// 32 keys * 256 B = 8192.
-#define TA_GEN6_QKT_PAIR(OFFBYTES, sj0, sj1, kr0, kr1, ka, qa, qb)          \\
-  asm volatile(                                                             \\
-      "{qkt}"                                                               \\
-      : "=&v"(sj0), "=&v"(sj1), "+&v"(kr0), "+&v"(kr1)                          \\
-      : "v"(ka[0]), "v"(ka[1]), "v"(ka[2]), "v"(ka[3]), "v"(ka[4]),         \\
-        "v"(ka[5]), "v"(ka[6]), "v"(ka[7]),                                 \\
-        "a"(qa[0]), "a"(qa[1]), "a"(qa[2]), "a"(qa[3]), "a"(qa[4]),         \\
-        "a"(qa[5]), "a"(qa[6]), "a"(qa[7]),                                 \\
-        "a"(qb[0]), "a"(qb[1]), "a"(qb[2]), "a"(qb[3]), "a"(qb[4]),         \\
-        "a"(qb[5]), "a"(qb[6]), "a"(qb[7]))
-'''
-    # OFF immediate substitution happens at expansion sites: we emit one
-    # macro per nb offset instead (asm cannot take immediates as operands
-    # for ds offsets via %N).
+def emit_probe_macros():
+    qkt = "\\n\\t".join(probe_qkt_pair_stream())
     bodies = []
     for nb in range(4):
         body = qkt.replace("offset:OFF", f"offset:{nb * 8192}")
@@ -153,7 +485,7 @@ def emit():
         "a"(qb[0]), "a"(qb[1]), "a"(qb[2]), "a"(qb[3]), "a"(qb[4]),         \\
         "a"(qb[5]), "a"(qb[6]), "a"(qb[7]))
 ''')
-    sm = "\\n\\t".join(softmax_pack_stream())
+    sm = "\\n\\t".join(probe_softmax_pack_stream())
     bodies.append(f'''#define TA_GEN6_SOFTMAX_PACK(c, alpha, m_run, l_run, sv, cl2)             \\
   asm volatile(                                                             \\
       "{sm}"                                                                \\
@@ -167,13 +499,35 @@ def emit():
         "v"(sv[10]), "v"(sv[11]), "v"(sv[12]), "v"(sv[13]), "v"(sv[14]),    \\
         "v"(sv[15]), "v"(cl2))
 ''')
-    src = "// GENERATED by tools/gen/gen_prefill6.py — do not edit by hand.\n#pragma once\n\n" + "\n".join(bodies)
+    return "\n".join(bodies)
+
+
+def stats(items):
+    from collections import Counter
+    c = Counter(k for _, k in items)
+    return dict(c)
+
+
+def emit():
+    parts = ["// GENERATED by tools/gen/gen_prefill6.py — do not edit by hand.",
+             "#pragma once", ""]
+    parts.append(emit_probe_macros())
+    parts.append(zero_o_macro())
+    parts.append(read_o_macros())
+    v1 = tile_body_v1()
+    v2 = tile_body_v2()
+    parts.append(f"// v1 (sequential phases): {stats(v1)}")
+    parts.append(tile_macro("TA_GEN6_TILE_V1", render(v1)))
+    parts.append(f"// v2 (staggered schedule): {stats(v2)}")
+    parts.append(tile_macro("TA_GEN6_TILE_V2", render(v2)))
+    src = "\n".join(parts)
     with open(OUT, "w") as f:
         f.write(src)
     print("wrote", OUT, len(src), "bytes")
+    print("v1:", stats(v1), " v2:", stats(v2))
     # the ninja hip rule has no header depfile and hipify caches on content:
     # force the next build to recompile the kernel TU
-    hipdir = os.path.dirname(OUT)
+    hipdir = os.path.dirname(os.path.abspath(OUT))
     for stale in ("fa_kernels_hip.cuda.o", "fa_kernels_hip.hip"):
         f = os.path.join(hipdir, stale)
         if os.path.exists(f):
