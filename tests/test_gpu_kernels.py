@@ -403,10 +403,16 @@ def test_combine_rescale_finish_kernels(ext):
     torch.testing.assert_close(lse_k.cpu(), lse_e, rtol=1e-5, atol=1e-5)
 
 
-@pytest.mark.parametrize("flag", ["TREE_ATTN_PREFILL4", "TREE_ATTN_PREFILL5"])
-def test_experimental_prefill_variants(ext, flag):
-    """fa_prefill4/5 (env-gated experiments) must stay numerically honest:
-    run in a subprocess because the route flag is latched at first call."""
+@pytest.mark.parametrize("flag,val", [
+    ("TREE_ATTN_PREFILL4", "1"), ("TREE_ATTN_PREFILL5", "1"),
+    ("TREE_ATTN_PREFILL6", "1"), ("TREE_ATTN_PREFILL6", "2"),
+])
+def test_experimental_prefill_variants(ext, flag, val):
+    """fa_prefill4/5/6 (env-gated variants) must stay numerically honest:
+    run in a subprocess because the route flag is latched at first call.
+    The shape list exercises gen6's asm interior loop (full 128-key tiles
+    below the causal frontier), the C boundary path (frontier, tails,
+    ragged rows) and the hand-off between them, plus GQA head mapping."""
     import os
     import subprocess
     import sys
@@ -417,19 +423,23 @@ def test_experimental_prefill_variants(ext, flag):
         "from tree_attention_torch_amd.ops.reference import flash_res_lse\n"
         "ext = flash._load_extension()\n"
         "scale = 128 ** -0.5\n"
-        "for tq, tkv, causal, h in [(256, 256, True, 4), (300, 300, True, 4),\n"
-        "                           (512, 1024, False, 2)]:\n"
+        "cases = [(256, 256, True, 4, 4), (300, 300, True, 4, 4),\n"
+        "         (512, 1024, False, 2, 2), (1024, 1024, True, 2, 2),\n"
+        "         (2048, 4096, True, 2, 2), (512, 512, True, 8, 2),\n"
+        "         (256, 320, False, 2, 2)]\n"  # ragged 64-key KV tail
+
+        "for tq, tkv, causal, hq, hkv in cases:\n"
         "    torch.manual_seed(1)\n"
-        "    q = torch.randn(1, h, tq, 128, device='cuda').bfloat16()\n"
-        "    k = torch.randn(1, h, tkv, 128, device='cuda').bfloat16()\n"
-        "    v = torch.randn(1, h, tkv, 128, device='cuda').bfloat16()\n"
+        "    q = torch.randn(1, hq, tq, 128, device='cuda').bfloat16()\n"
+        "    k = torch.randn(1, hkv, tkv, 128, device='cuda').bfloat16()\n"
+        "    v = torch.randn(1, hkv, tkv, 128, device='cuda').bfloat16()\n"
         "    o, l = ext.flash_attention(q, k, v, scale, causal, tkv - tq, 0)\n"
         "    ro, rl = flash_res_lse(q.cpu(), k.cpu(), v.cpu(), scale, causal, tkv - tq, 0)\n"
         "    torch.testing.assert_close(o.cpu(), ro, rtol=2.5e-2, atol=2.5e-2)\n"
         "    torch.testing.assert_close(l.cpu(), rl, rtol=1e-3, atol=1e-3)\n"
         "print('VARIANT_OK')\n"
     )
-    env = dict(os.environ, **{flag: "1"})
+    env = dict(os.environ, **{flag: val})
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, env=env, timeout=300)
     assert r.returncode == 0 and "VARIANT_OK" in r.stdout, r.stdout + r.stderr
