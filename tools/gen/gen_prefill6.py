@@ -58,7 +58,18 @@ OP_L = {0: "%2", 1: "%3"}
 OP_KA = lambda s: f"%{4 + s}"          # noqa: E731
 OP_VA = lambda c, nd: f"%{12 + c * 4 + nd}"   # noqa: E731
 OP_CL2 = "%20"
-OP_Q = lambda j, s: f"%{21 + j * 8 + s}"      # noqa: E731
+# Q fragments live in literal AGPRs a[128:191], written once by the
+# prologue LOAD_Q macros. Passing them as 16 "a" operands instead made
+# the register allocator REMATERIALIZE the global loads inside every tile
+# iteration, and hipcc brackets each reload with s_waitcnt vmcnt(0) —
+# draining the glds staging pipeline 8x per tile (measured: gen6 at
+# prefill2 parity instead of past it; guide §5 ".s-level trap (b)").
+QBASE = 128
+
+
+def OP_Q(j, s):
+    b = QBASE + j * 32 + s * 4
+    return f"a[{b}:{b + 3}]"
 
 
 def vr(base, n=4):
@@ -340,7 +351,7 @@ def render(body_items):
 
 
 def clobbers():
-    regs = [f'"a{i}"' for i in range(128)]
+    regs = [f'"a{i}"' for i in range(192)]  # O a[0:127] + Q a[128:191]
     regs += [f'"v{i}"' for i in range(144, 256)]
     regs += ['"s40"', '"s41"']  # s_call_b64 return address
     regs += ['"vcc"', '"memory"']
@@ -353,7 +364,7 @@ def tile_macro(name, body):
     barrier, 4 nb phases, shared rescale subroutines, tail barrier.
     Labels get a per-expansion suffix via %=."""
     body = body.replace("TA_G6_RS", "TA_G6_%=_RS")
-    return f'''#define {name}(m0, m1, l0, l1, ka, va, cl2, qa, qb)                 \\
+    return f'''#define {name}(m0, m1, l0, l1, ka, va, cl2)                        \\
   asm volatile(                                                             \\
       "s_waitcnt vmcnt(16)\\n\\t"                                            \\
       "s_barrier\\n\\t"                                                      \\
@@ -364,13 +375,28 @@ def tile_macro(name, body):
         "v"(ka[5]), "v"(ka[6]), "v"(ka[7]),                                 \\
         "v"(va[0]), "v"(va[1]), "v"(va[2]), "v"(va[3]), "v"(va[4]),         \\
         "v"(va[5]), "v"(va[6]), "v"(va[7]),                                 \\
-        "v"(cl2),                                                           \\
-        "a"(qa[0]), "a"(qa[1]), "a"(qa[2]), "a"(qa[3]), "a"(qa[4]),         \\
-        "a"(qa[5]), "a"(qa[6]), "a"(qa[7]),                                 \\
-        "a"(qb[0]), "a"(qb[1]), "a"(qb[2]), "a"(qb[3]), "a"(qb[4]),         \\
-        "a"(qb[5]), "a"(qb[6]), "a"(qb[7])                                  \\
+        "v"(cl2)                                                            \\
       : {clobbers()})
 '''
+
+
+def load_q_macros():
+    """Prologue: write one Q V8 slice (4 dwords) into its literal AGPRs.
+    The s_nop 1 covers the v_accvgpr_write -> MFMA-operand hazard (the
+    first QKT is a barrier away, but the guard is free here)."""
+    out = []
+    for j in range(2):
+        for s in range(8):
+            b = QBASE + j * 32 + s * 4
+            writes = "\\n\\t".join(
+                f"v_accvgpr_write_b32 a{b + u}, %{u}" for u in range(4))
+            clob = ", ".join(f'"a{b + u}"' for u in range(4))
+            out.append(
+                f'''#define TA_GEN6_LOAD_Q_J{j}_S{s}(w0, w1, w2, w3)                        \\
+  asm volatile("{writes}\\n\\ts_nop 1"                                       \\
+               :: "v"(w0), "v"(w1), "v"(w2), "v"(w3) : {clob})
+''')
+    return "\n".join(out)
 
 
 def zero_o_macro():
@@ -513,6 +539,7 @@ def emit():
              "#pragma once", ""]
     parts.append(emit_probe_macros())
     parts.append(zero_o_macro())
+    parts.append(load_q_macros())
     parts.append(read_o_macros())
     v1 = tile_body_v1()
     v2 = tile_body_v2()
