@@ -51,6 +51,14 @@ T0, T1 = "v168", "v169"
 MT = {0: "v170", 1: "v171"}
 RS, P0, P1 = "v172", "v173", "v174"
 AL = {0: "v156", 1: "v157"}
+# zipped-stream per-block temp sets (VERDICT: at 1 wave/SIMD nothing hides
+# dependent-VALU latency, so the two blocks' softmax chains interleave at
+# instruction granularity — consecutive ops independent). Block 1's p1/rs
+# borrow v144/v145 from the VR pool: softmax fills run in QKT phases, the
+# V-fragment slots only in PV phases — lifetimes disjoint.
+TT = {0: ("v168", "v169"), 1: ("v158", "v159")}   # permlane temp pairs
+PP = {0: ("v172", "v173"), 1: ("v175", "v144")}   # p0/p1 per block
+RSJ = {0: "v174", 1: "v145"}                      # rowsum per block
 
 # tile-statement operand indices
 OP_M = {0: "%0", 1: "%1"}
@@ -177,6 +185,91 @@ def sm_finish_stream(j, buf):
     return L
 
 
+def zip_streams(a, b):
+    """Interleave two independent instruction streams one-for-one (nops
+    and labels pass through without pairing)."""
+    out = []
+    ia = ib = 0
+    while ia < len(a) or ib < len(b):
+        if ia < len(a):
+            out.append(a[ia])
+            ia += 1
+        if ib < len(b):
+            out.append(b[ib])
+            ib += 1
+    return out
+
+
+def _sm_start_one(j, buf):
+    """One block's startSM with per-block temps (for zipping)."""
+    s = lambda i: sj_reg(buf, j, i)  # noqa: E731
+    t0, t1 = TT[j]
+    mt, al, m = MT[j], AL[j], OP_M[j]
+    return [
+        (f"v_max3_f32 {mt}, {s(0)}, {s(1)}, {s(2)}", "valu"),
+        (f"v_max3_f32 {t0}, {s(3)}, {s(4)}, {s(5)}", "valu"),
+        (f"v_max3_f32 {t1}, {s(6)}, {s(7)}, {s(8)}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {t0}, {t1}", "valu"),
+        (f"v_max3_f32 {t0}, {s(9)}, {s(10)}, {s(11)}", "valu"),
+        (f"v_max3_f32 {t1}, {s(12)}, {s(13)}, {s(14)}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {t0}, {t1}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {s(15)}, {s(15)}", "valu"),
+        (f"v_mov_b32 {t0}, {mt}", "valu"),
+        (f"v_mov_b32 {t1}, {mt}", "valu"),
+        ("s_nop 0", "nop"),  # VGPR write -> v_permlane (zip partner covers 1)
+        (f"v_permlane32_swap_b32 {t0}, {t1}", "valu"),
+        (f"v_max_f32 {mt}, {t0}, {t1}", "valu"),
+        (f"v_mul_f32 {mt}, {mt}, {OP_CL2}", "valu"),
+        (f"v_max_f32 {mt}, {m}, {mt}", "valu"),
+        (f"v_sub_f32 {t0}, {m}, {mt}", "valu"),
+        (f"v_exp_f32 {al}, {t0}", "trans"),
+        (f"v_mov_b32 {m}, {mt}", "valu"),
+    ]
+
+
+def _sm_finish_one(j, buf):
+    """One block's finishSM with per-block temps (for zipping)."""
+    s = lambda i: sj_reg(buf, j, i)  # noqa: E731
+    p0, p1 = PP[j]
+    rs = RSJ[j]
+    mt, al, l = MT[j], AL[j], OP_L[j]
+    c = lambda i: f"v{CP[j] + i}"  # noqa: E731
+    L = [(f"v_mov_b32 {rs}, 0", "valu")]
+    for i in range(8):
+        L += [
+            (f"v_fma_f32 {p0}, {s(2 * i)}, {OP_CL2}, -{mt}", "valu"),
+            (f"v_fma_f32 {p1}, {s(2 * i + 1)}, {OP_CL2}, -{mt}", "valu"),
+            (f"v_exp_f32 {p0}, {p0}", "trans"),
+            (f"v_exp_f32 {p1}, {p1}", "trans"),
+            # TRANS->reader 1 state: the zip partner's op covers it
+            (f"v_cvt_pk_bf16_f32 {c(i)}, {p0}, {p1}", "valu"),
+            (f"v_add_f32 {p0}, {p0}, {p1}", "valu"),
+            (f"v_add_f32 {rs}, {rs}, {p0}", "valu"),
+        ]
+    L += [
+        (f"v_mov_b32 {p0}, {rs}", "valu"),
+        (f"v_mov_b32 {p1}, {rs}", "valu"),
+        ("s_nop 0", "nop"),
+        (f"v_permlane32_swap_b32 {p0}, {p1}", "valu"),
+        (f"v_add_f32 {rs}, {p0}, {p1}", "valu"),
+        (f"v_fma_f32 {l}, {l}, {al}, {rs}", "valu"),
+        ("s_nop 0", "nop"),
+        (f"v_permlane32_swap_b32 {c(0)}, {c(2)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(1)}, {c(3)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(4)}, {c(6)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(5)}, {c(7)}", "valu"),
+    ]
+    return L
+
+
+def sm_zip_start(buf):
+    return zip_streams(_sm_start_one(0, buf), _sm_start_one(1, buf))
+
+
+def sm_zip_finish(buf):
+    return zip_streams(_sm_finish_one(0, buf), _sm_finish_one(1, buf))
+
+
 def rescale_call(j, label):
     """Vote-skipped O rescale call site: if any lane's alpha != 1, call the
     per-statement shared rescale subroutine for block j (the 200-instr
@@ -266,10 +359,8 @@ def tile_body_v1():
     for nb in range(4):
         L += qkt_stream(nb, "A")
         L.append(("s_nop 11", "nop"))   # QKT D -> softmax VALU reader
-        L += sm_start_stream(0, "A")
-        L += sm_finish_stream(0, "A")
-        L += sm_start_stream(1, "A")
-        L += sm_finish_stream(1, "A")
+        L += sm_zip_start("A")
+        L += sm_zip_finish("A")
         L += rescale_call(0, f"V1N{nb}")
         L += rescale_call(1, f"V1N{nb}")
         L += pv_stream(nb)
@@ -311,8 +402,7 @@ def tile_body_v2():
     # ramp: QKT(0) -> sjA; QKT(1) -> sjB with SM(0) as fills
     L += qkt_stream(0, "A")
     L.append(("s_nop 11", "nop"))
-    sm0 = sm_start_stream(0, "A") + sm_finish_stream(0, "A") + \
-        sm_start_stream(1, "A") + sm_finish_stream(1, "A")
+    sm0 = sm_zip_start("A") + sm_zip_finish("A")
     L += interleave(qkt_stream(1, "B"), sm0)
     L += rescale_call(0, "V2R") + rescale_call(1, "V2R")
     # steady state over nb = 1..3
@@ -320,11 +410,11 @@ def tile_body_v2():
         cur = "B" if (nb % 2) else "A"    # sj buffer holding scores(nb)
         nxt = "A" if (nb % 2) else "B"
         # phase X: PV(nb-1) || startSM(nb) both blocks
-        startf = sm_start_stream(0, cur) + sm_start_stream(1, cur)
+        startf = sm_zip_start(cur)
         L += interleave(pv_stream(nb - 1), startf)
         L.append(("s_nop 11", "nop"))
         # phase Y: QKT(nb+1) || finishSM(nb); last nb has no QKT(4)
-        finishf = sm_finish_stream(0, cur) + sm_finish_stream(1, cur)
+        finishf = sm_zip_finish(cur)
         if nb < 3:
             L += interleave(qkt_stream(nb + 1, nxt), finishf)
         else:
