@@ -618,6 +618,30 @@ def emit_probe_macros():
     return "\n".join(bodies)
 
 
+def tile_body_ablate(which):
+    """Timing-ablation tile bodies (guide §5 common-mistake 8: ablate
+    before optimizing). Outputs are garbage — tools-only, env-gated.
+    which: "qkt" = staging skeleton + QKT streams only;
+           "qktsm" = + zipped softmax; "qktpv" = QKT + PV with constant
+           P packs (softmax skipped; c = 1.0 pairs so MFMA data is sane
+           for DVFS comparability)."""
+    L = []
+    for nb in range(4):
+        L += qkt_stream(nb, "A")
+        L.append(("s_nop 11", "nop"))
+        if which == "qktsm":
+            L += sm_zip_start("A")
+            L += sm_zip_finish("A")
+        elif which == "qktpv":
+            if nb == 0:
+                for j in range(2):
+                    for i in range(8):
+                        L.append((f"v_mov_b32 v{CP[j] + i}, 0x3f803f80",
+                                  "valu"))
+            L += pv_stream(nb)
+    return L
+
+
 def stats(items):
     from collections import Counter
     c = Counter(k for _, k in items)
@@ -637,6 +661,10 @@ def emit():
     parts.append(tile_macro("TA_GEN6_TILE_V1", render(v1)))
     parts.append(f"// v2 (staggered schedule): {stats(v2)}")
     parts.append(tile_macro("TA_GEN6_TILE_V2", render(v2)))
+    for ver, which in ((3, "qkt"), (4, "qktsm"), (5, "qktpv")):
+        body = tile_body_ablate(which)
+        parts.append(f"// ablation {which}: {stats(body)}")
+        parts.append(tile_macro(f"TA_GEN6_TILE_V{ver}", render(body)))
     src = "\n".join(parts)
     with open(OUT, "w") as f:
         f.write(src)

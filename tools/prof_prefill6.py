@@ -20,7 +20,15 @@ torch.manual_seed(0)
 q = torch.randn(1, 32, tq, 128, device="cuda").bfloat16()
 k = torch.randn(1, 32, tkv, 128, device="cuda").bfloat16()
 v = torch.randn(1, 32, tkv, 128, device="cuda").bfloat16()
+for _ in range(3):
+    local_attention(q, k, v, is_causal=causal, q_offset=tkv - tq)
+torch.cuda.synchronize()
+import time  # noqa: E402
+
+t0 = time.perf_counter()
 for _ in range(iters):
     local_attention(q, k, v, is_causal=causal, q_offset=tkv - tq)
 torch.cuda.synchronize()
-print("done")
+dt = (time.perf_counter() - t0) / iters
+pairs = tq * (tkv - tq) + tq * (tq + 1) // 2 if causal else tq * tkv
+print("%.3f ms  %.1f TF/s" % (dt * 1e3, 2 * 2 * 32 * pairs * 128 / dt / 1e12))
