@@ -406,6 +406,7 @@ def test_combine_rescale_finish_kernels(ext):
 @pytest.mark.parametrize("flag,val", [
     ("TREE_ATTN_PREFILL4", "1"), ("TREE_ATTN_PREFILL5", "1"),
     ("TREE_ATTN_PREFILL6", "1"), ("TREE_ATTN_PREFILL6", "2"),
+    ("TREE_ATTN_PREFILL6", "7"),
 ])
 def test_experimental_prefill_variants(ext, flag, val):
     """fa_prefill4/5/6 (env-gated variants) must stay numerically honest:

@@ -618,6 +618,247 @@ def emit_probe_macros():
     return "\n".join(bodies)
 
 
+# ---------------------------------------------------------------------------
+# gen7: the occupancy-2 variant. 8 waves (2/SIMD), ONE 32-row q-block per
+# wave — the co-resident partner wave hides dependent-VALU latency and
+# MFMA-RAW stalls that the 1-wave/SIMD schedule pays in wall time (measured:
+# v2's wall = 8.0k active + 6.3k stall cycles per tile; total per-SIMD issue
+# work is the same at occ 2, so the ceiling is the ~8k issue-bound floor).
+#
+# Register budget at __launch_bounds__(512, 2): 256 per wave, and the
+# compiler reserves arch VGPRs >= v128 (measured warning), so every fixed
+# arch register sits in v[96:127]: scores v[96:111], c packs v[112:119],
+# kr v[120:127] in QKT phases = vr (two 4-reg fragment slots) in PV phases
+# (disjoint lifetimes). Temps are operand-allocated scratch ("=&v") from
+# the compiler's v0..v95 pool (scalar f32 operands, so %N names work).
+# O a[0:63], Q a[64:95]. The LDS double buffer folds into the ds-offset
+# immediates (+32768 for buffer 1: max base+imm < 64 KiB fits the 16-bit
+# field), so ONE ka/va address set serves both buffers.
+# ---------------------------------------------------------------------------
+
+G7_SJ = 96
+G7_CP = 112
+G7_KR = {0: 120, 1: 124}
+G7_VRSLOT = [120, 124]
+# operands: outputs first. %0 m, %1 l, %2..%8 temps; inputs %9..%16 ka,
+# %17..%24 va, %25 cl2
+G7_M, G7_L = "%0", "%1"
+G7_T0, G7_T1, G7_MT, G7_RS, G7_P0, G7_P1, G7_AL = (
+    "%2", "%3", "%4", "%5", "%6", "%7", "%8")
+G7_KA = lambda s: f"%{9 + s}"              # noqa: E731
+G7_VA = lambda c, nd: f"%{17 + c * 4 + nd}"   # noqa: E731
+G7_CL2 = "%25"
+
+
+def g7_sj(i):
+    return f"v{G7_SJ + i}"
+
+
+def g7_qkt_stream(nb, bufoff):
+    """Single-chain QK^T: 8 ds_read_b128 (one ahead) + 8 MFMAs, D-as-C."""
+    off = bufoff + nb * 8192
+    s0 = f"v[{G7_SJ}:{G7_SJ + 15}]"
+    kr0, kr1 = vr(G7_KR[0]), vr(G7_KR[1])
+    qreg = lambda s: f"a[{64 + s * 4}:{64 + s * 4 + 3}]"  # noqa: E731
+    L = []
+    L.append((f"ds_read_b128 {kr0}, {G7_KA(0)} offset:{off}", "ds"))
+    L.append((f"ds_read_b128 {kr1}, {G7_KA(1)} offset:{off}", "ds"))
+    L.append(("s_waitcnt lgkmcnt(1)", "wait"))
+    L.append((f"{MFMA} {s0}, {kr0}, {qreg(0)}, 0", "mfma"))
+    for s in range(1, 8):
+        cur = kr1 if (s % 2) else kr0
+        nxt = kr0 if (s % 2) else kr1
+        if s + 1 < 8:
+            L.append((f"ds_read_b128 {nxt}, {G7_KA(s + 1)} offset:{off}", "ds"))
+            L.append(("s_waitcnt lgkmcnt(1)", "wait"))
+        else:
+            L.append(("s_waitcnt lgkmcnt(0)", "wait"))
+        L.append((f"{MFMA} {s0}, {cur}, {qreg(s)}, {s0}", "mfma"))
+    return L
+
+
+def g7_sm_stream():
+    """Single-block softmax (start+finish), exp2 domain."""
+    s = g7_sj
+    t0, t1, mt, rs, p0, p1, al = (G7_T0, G7_T1, G7_MT, G7_RS, G7_P0,
+                                  G7_P1, G7_AL)
+    m, l, cl = G7_M, G7_L, G7_CL2
+    c = lambda i: f"v{G7_CP + i}"  # noqa: E731
+    L = [
+        (f"v_max3_f32 {mt}, {s(0)}, {s(1)}, {s(2)}", "valu"),
+        (f"v_max3_f32 {t0}, {s(3)}, {s(4)}, {s(5)}", "valu"),
+        (f"v_max3_f32 {t1}, {s(6)}, {s(7)}, {s(8)}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {t0}, {t1}", "valu"),
+        (f"v_max3_f32 {t0}, {s(9)}, {s(10)}, {s(11)}", "valu"),
+        (f"v_max3_f32 {t1}, {s(12)}, {s(13)}, {s(14)}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {t0}, {t1}", "valu"),
+        (f"v_max3_f32 {mt}, {mt}, {s(15)}, {s(15)}", "valu"),
+        (f"v_mov_b32 {t0}, {mt}", "valu"),
+        (f"v_mov_b32 {t1}, {mt}", "valu"),
+        ("s_nop 1", "nop"),
+        (f"v_permlane32_swap_b32 {t0}, {t1}", "valu"),
+        (f"v_max_f32 {mt}, {t0}, {t1}", "valu"),
+        (f"v_mul_f32 {mt}, {mt}, {cl}", "valu"),
+        (f"v_max_f32 {mt}, {m}, {mt}", "valu"),
+        (f"v_sub_f32 {t0}, {m}, {mt}", "valu"),
+        (f"v_exp_f32 {al}, {t0}", "trans"),
+        (f"v_mov_b32 {m}, {mt}", "valu"),
+        (f"v_mov_b32 {rs}, 0", "valu"),
+    ]
+    for i in range(8):
+        L += [
+            (f"v_fma_f32 {p0}, {s(2 * i)}, {cl}, -{mt}", "valu"),
+            (f"v_fma_f32 {p1}, {s(2 * i + 1)}, {cl}, -{mt}", "valu"),
+            (f"v_exp_f32 {p0}, {p0}", "trans"),
+            (f"v_exp_f32 {p1}, {p1}", "trans"),
+            ("s_nop 0", "nop"),
+            (f"v_cvt_pk_bf16_f32 {c(i)}, {p0}, {p1}", "valu"),
+            (f"v_add_f32 {p0}, {p0}, {p1}", "valu"),
+            (f"v_add_f32 {rs}, {rs}, {p0}", "valu"),
+        ]
+    L += [
+        (f"v_mov_b32 {t0}, {rs}", "valu"),
+        (f"v_mov_b32 {t1}, {rs}", "valu"),
+        ("s_nop 1", "nop"),
+        (f"v_permlane32_swap_b32 {t0}, {t1}", "valu"),
+        (f"v_add_f32 {rs}, {t0}, {t1}", "valu"),
+        (f"v_fma_f32 {l}, {l}, {al}, {rs}", "valu"),
+        ("s_nop 1", "nop"),
+        (f"v_permlane32_swap_b32 {c(0)}, {c(2)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(1)}, {c(3)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(4)}, {c(6)}", "valu"),
+        (f"v_permlane32_swap_b32 {c(5)}, {c(7)}", "valu"),
+    ]
+    return L
+
+
+def g7_rescale_call(label):
+    return [
+        (f"v_cmp_neq_f32 vcc, 1.0, {G7_AL}", "valu"),
+        ("s_nop 0", "nop"),
+        (f"s_cbranch_vccz TA_G6_RSD{label}", "salu"),
+        (f"s_call_b64 s[40:41], TA_G6_RSFN7", "salu"),
+        (f"TA_G6_RSD{label}:", "label"),
+    ]
+
+
+def g7_rescale_tail():
+    t = [G7_T0, G7_T1, G7_MT, G7_RS]
+    L = [("s_branch TA_G6_RSEND", "salu"),
+         ("TA_G6_RSFN7:", "label"),
+         ("s_nop 11", "nop")]
+    for r in range(0, 64, 4):
+        for u in range(4):
+            L.append((f"v_accvgpr_read_b32 {t[u]}, a{r + u}", "valu"))
+        for u in range(4):
+            L.append((f"v_mul_f32 {t[u]}, {t[u]}, {G7_AL}", "valu"))
+        for u in range(4):
+            L.append((f"v_accvgpr_write_b32 a{r + u}, {t[u]}", "valu"))
+    L.append(("s_nop 1", "nop"))
+    L.append(("s_setpc_b64 s[40:41]", "salu"))
+    L.append(("TA_G6_RSEND:", "label"))
+    return L
+
+
+def g7_pv_stream(nb, bufoff):
+    """PV: 16 tr-reads (two 4-reg slots, one fragment of lookahead) +
+    8 MFMAs into a[0:63]. The slots reuse kr's v[120:127] (QKT-only)."""
+    L = []
+    frags = [(ks, nd) for ks in range(2) for nd in range(4)]
+
+    def emit_reads(fi):
+        ks, nd = frags[fi]
+        slot = G7_VRSLOT[fi % 2]
+        off = bufoff + nb * 8192 + ks * 4096
+        L.append((f"ds_read_b64_tr_b16 {vr(slot, 2)}, {G7_VA(0, nd)} offset:{off}", "ds"))
+        L.append((f"ds_read_b64_tr_b16 {vr(slot + 2, 2)}, {G7_VA(1, nd)} offset:{off}", "ds"))
+
+    emit_reads(0)
+    for fi in range(8):
+        ks, nd = frags[fi]
+        slot = G7_VRSLOT[fi % 2]
+        if fi + 1 < 8:
+            emit_reads(fi + 1)
+            L.append(("s_waitcnt lgkmcnt(2)", "wait"))
+        else:
+            L.append(("s_waitcnt lgkmcnt(0)", "wait"))
+        a = vr(slot)
+        L.append((f"{MFMA} a[{nd * 16}:{nd * 16 + 15}], {a}, "
+                  f"v[{G7_CP + ks * 4}:{G7_CP + ks * 4 + 3}], "
+                  f"a[{nd * 16}:{nd * 16 + 15}]", "mfma"))
+    return L
+
+
+def g7_tile_body(bufoff):
+    L = []
+    for nb in range(4):
+        L += g7_qkt_stream(nb, bufoff)
+        L.append(("s_nop 11", "nop"))   # QKT D -> softmax VALU reader
+        L += g7_sm_stream()
+        L += g7_rescale_call(f"N{nb}")
+        L += g7_pv_stream(nb, bufoff)
+        if nb < 3:
+            L.append(("s_nop 11", "nop"))
+    L += g7_rescale_tail()
+    return L
+
+
+def g7_clobbers():
+    regs = [f'"a{i}"' for i in range(96)]   # O a[0:63] + Q a[64:95]
+    regs += [f'"v{i}"' for i in range(96, 128)]
+    regs += ['"s40"', '"s41"', '"vcc"', '"memory"']
+    return ", ".join(regs)
+
+
+def g7_tile_macro(name, body):
+    body = body.replace("TA_G6_RS", "TA_G6_%=_RS")
+    return (
+        f"#define {name}(m0, l0, tmp, ka, va, cl2)                          \\\n"
+        "  asm volatile(                                                    \\\n"
+        '      "s_waitcnt vmcnt(8)\\n\\t"                                    \\\n'
+        '      "s_barrier\\n\\t"                                             \\\n'
+        f'      "{body}\\n\\t"                                               \\\n'
+        '      "s_barrier"                                                  \\\n'
+        '      : "+v"(m0), "+v"(l0),                                        \\\n'
+        '        "=&v"(tmp[0]), "=&v"(tmp[1]), "=&v"(tmp[2]), "=&v"(tmp[3]),\\\n'
+        '        "=&v"(tmp[4]), "=&v"(tmp[5]), "=&v"(tmp[6])                \\\n'
+        '      : "v"(ka[0]), "v"(ka[1]), "v"(ka[2]), "v"(ka[3]), "v"(ka[4]),\\\n'
+        '        "v"(ka[5]), "v"(ka[6]), "v"(ka[7]),                        \\\n'
+        '        "v"(va[0]), "v"(va[1]), "v"(va[2]), "v"(va[3]), "v"(va[4]),\\\n'
+        '        "v"(va[5]), "v"(va[6]), "v"(va[7]),                        \\\n'
+        '        "v"(cl2)                                                   \\\n'
+        f"      : {g7_clobbers()})\n"
+    )
+
+
+def g7_aux_macros():
+    parts = []
+    writes = "\\n\\t".join(
+        f"v_accvgpr_write_b32 a{i}, 0" for i in range(64))
+    aclob = ", ".join(f'"a{i}"' for i in range(64))
+    parts.append(
+        "#define TA_GEN7_ZERO_O()                                         \\\n"
+        f'  asm volatile("{writes}" ::: {aclob})\n')
+    for s in range(8):
+        b = 64 + s * 4
+        w = "\\n\\t".join(
+            f"v_accvgpr_write_b32 a{b + u}, %{u}" for u in range(4))
+        clob = ", ".join(f'"a{b + u}"' for u in range(4))
+        parts.append(
+            f"#define TA_GEN7_LOAD_Q_S{s}(w0, w1, w2, w3)                  \\\n"
+            f'  asm volatile("{w}\\n\\ts_nop 1"                             \\\n'
+            f'               :: "v"(w0), "v"(w1), "v"(w2), "v"(w3) : {clob})\n')
+    for nd in range(4):
+        base = nd * 16
+        reads = "\\n\\t".join(
+            f"v_accvgpr_read_b32 %{i}, a{base + i}" for i in range(16))
+        outs = ", ".join(f'"=v"(f[{i}])' for i in range(16))
+        parts.append(
+            f"#define TA_GEN7_READ_O_ND{nd}(f)                             \\\n"
+            f'  asm volatile("s_nop 11\\n\\t{reads}" : {outs})\n')
+    return "\n".join(parts)
+
+
 def tile_body_ablate(which):
     """Timing-ablation tile bodies (guide §5 common-mistake 8: ablate
     before optimizing). Outputs are garbage — tools-only, env-gated.
@@ -665,6 +906,11 @@ def emit():
         body = tile_body_ablate(which)
         parts.append(f"// ablation {which}: {stats(body)}")
         parts.append(tile_macro(f"TA_GEN6_TILE_V{ver}", render(body)))
+    parts.append(g7_aux_macros())
+    for buf in (0, 1):
+        g7 = g7_tile_body(buf * 32768)
+        parts.append(f"// gen7 buf{buf} (occ 2, one q-block/wave): {stats(g7)}")
+        parts.append(g7_tile_macro(f"TA_GEN7_TILE_B{buf}", render(g7)))
     src = "\n".join(parts)
     with open(OUT, "w") as f:
         f.write(src)
