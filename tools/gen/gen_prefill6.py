@@ -639,7 +639,7 @@ def emit_probe_macros():
 G7_SJ = 96
 G7_CP = 112
 G7_KR = {0: 120, 1: 124}
-G7_VRSLOT = [120, 124]
+G7_VRSLOT = [120, 124, 96]  # third slot borrows score regs (dead in PV)
 # operands: outputs first. %0 m, %1 l, %2..%8 temps; inputs %9..%16 ka,
 # %17..%24 va, %25 cl2
 G7_M, G7_L = "%0", "%1"
@@ -655,25 +655,25 @@ def g7_sj(i):
 
 
 def g7_qkt_stream(nb, bufoff):
-    """Single-chain QK^T: 8 ds_read_b128 (one ahead) + 8 MFMAs, D-as-C."""
+    """Single-chain QK^T: 8 ds_read_b128 (TWO ahead: LDS dependent-read
+    latency ~50 cyc > one 32-cyc MFMA; the third buffer borrows the c-pack
+    registers v[112:115], dead during QKT) + 8 MFMAs, D-as-C."""
     off = bufoff + nb * 8192
     s0 = f"v[{G7_SJ}:{G7_SJ + 15}]"
-    kr0, kr1 = vr(G7_KR[0]), vr(G7_KR[1])
+    bufs = [vr(G7_KR[0]), vr(G7_KR[1]), vr(G7_CP)]
     qreg = lambda s: f"a[{64 + s * 4}:{64 + s * 4 + 3}]"  # noqa: E731
     L = []
-    L.append((f"ds_read_b128 {kr0}, {G7_KA(0)} offset:{off}", "ds"))
-    L.append((f"ds_read_b128 {kr1}, {G7_KA(1)} offset:{off}", "ds"))
-    L.append(("s_waitcnt lgkmcnt(1)", "wait"))
-    L.append((f"{MFMA} {s0}, {kr0}, {qreg(0)}, 0", "mfma"))
-    for s in range(1, 8):
-        cur = kr1 if (s % 2) else kr0
-        nxt = kr0 if (s % 2) else kr1
-        if s + 1 < 8:
-            L.append((f"ds_read_b128 {nxt}, {G7_KA(s + 1)} offset:{off}", "ds"))
-            L.append(("s_waitcnt lgkmcnt(1)", "wait"))
-        else:
-            L.append(("s_waitcnt lgkmcnt(0)", "wait"))
-        L.append((f"{MFMA} {s0}, {cur}, {qreg(s)}, {s0}", "mfma"))
+    L.append((f"ds_read_b128 {bufs[0]}, {G7_KA(0)} offset:{off}", "ds"))
+    L.append((f"ds_read_b128 {bufs[1]}, {G7_KA(1)} offset:{off}", "ds"))
+    L.append((f"ds_read_b128 {bufs[2]}, {G7_KA(2)} offset:{off}", "ds"))
+    for s in range(8):
+        # before MFMA(s): outstanding = reads for s, s+1, s+2 (minus the
+        # tail); wait to <= the younger two so read(s) has landed
+        L.append((f"s_waitcnt lgkmcnt({min(2, 7 - s)})", "wait"))
+        cacc = "0" if s == 0 else s0
+        L.append((f"{MFMA} {s0}, {bufs[s % 3]}, {qreg(s)}, {cacc}", "mfma"))
+        if s + 3 < 8:
+            L.append((f"ds_read_b128 {bufs[(s + 3) % 3]}, {G7_KA(s + 3)} offset:{off}", "ds"))
     return L
 
 
@@ -768,17 +768,20 @@ def g7_pv_stream(nb, bufoff):
 
     def emit_reads(fi):
         ks, nd = frags[fi]
-        slot = G7_VRSLOT[fi % 2]
+        slot = G7_VRSLOT[fi % 3]
         off = bufoff + nb * 8192 + ks * 4096
         L.append((f"ds_read_b64_tr_b16 {vr(slot, 2)}, {G7_VA(0, nd)} offset:{off}", "ds"))
         L.append((f"ds_read_b64_tr_b16 {vr(slot + 2, 2)}, {G7_VA(1, nd)} offset:{off}", "ds"))
 
     emit_reads(0)
+    emit_reads(1)
     for fi in range(8):
         ks, nd = frags[fi]
-        slot = G7_VRSLOT[fi % 2]
-        if fi + 1 < 8:
-            emit_reads(fi + 1)
+        slot = G7_VRSLOT[fi % 3]
+        if fi + 2 < 8:
+            emit_reads(fi + 2)
+            L.append(("s_waitcnt lgkmcnt(4)", "wait"))
+        elif fi + 2 == 8:
             L.append(("s_waitcnt lgkmcnt(2)", "wait"))
         else:
             L.append(("s_waitcnt lgkmcnt(0)", "wait"))
