@@ -914,6 +914,13 @@ def emit():
         g7 = g7_tile_body(buf * 32768)
         parts.append(f"// gen7 buf{buf} (occ 2, one q-block/wave): {stats(g7)}")
         parts.append(g7_tile_macro(f"TA_GEN7_TILE_B{buf}", render(g7)))
+    # no-sync ablation: same body, no vmcnt/barriers (garbage output) —
+    # isolates the per-tile barrier + DMA-wait share of the wall time
+    g8 = g7_tile_body(0)
+    parts.append(g7_tile_macro("TA_GEN7_TILE_NOSYNC", render(g8))
+                 .replace('"s_waitcnt vmcnt(8)\\n\\t"', '""')
+                 .replace('"s_barrier\\n\\t"', '""')
+                 .replace('"s_barrier"', '"s_nop 0"'))
     src = "\n".join(parts)
     with open(OUT, "w") as f:
         f.write(src)
