@@ -136,16 +136,32 @@ def local_attention(
                 f"Import error: {_EXT_ERR}"
             )
         d = q.shape[-1]
-        # D=64 is native for bf16/fp16 (decode and prefill kernels)
-        native_d64 = d == 64 and k.dtype in (torch.bfloat16, torch.float16)
-        if d < 128 and not native_d64:
-            # exact zero-padding fallback for narrow heads: padded dims add 0
-            # to every q.k score and the padded output columns are sliced off.
-            # (softmax_scale above was computed from the REAL D.) Native
-            # D<128 kernels are future work; this costs the padding copies.
+        halfish = k.dtype in (torch.bfloat16, torch.float16)
+        # D=64 is native for bf16/fp16 (decode and prefill kernels);
+        # D in {32,48,80,96,112} is native for bf16/fp16 DECODE (the
+        # zero-padded-LDS narrow-head kernel — no global pad copies).
+        native_d64 = d == 64 and halfish
+        g = q.shape[1] // k.shape[1]
+        tq = q.shape[2]
+        # will the extension run the prefill kernel (vs decode/spec-loop)?
+        will_prefill = tq * g > 16 and \
+            decode_loop_chunks(tq, g, q.shape[0], q.shape[1]) == 0
+        narrow_decode = (halfish and not will_prefill
+                         and d in (32, 48, 80, 96, 112))
+        if d < 128 and not native_d64 and not narrow_decode:
+            # exact zero-padding fallback: padded dims add 0 to every q.k
+            # score and the padded output columns are sliced off.
+            # (softmax_scale above was computed from the REAL D.) Pad to the
+            # SMALLEST natively supported dim, not always 128.
             import torch.nn.functional as F
 
-            pad = 128 - d
+            if halfish:
+                supported = ((64, 128) if will_prefill
+                             else (32, 48, 64, 80, 96, 112, 128))
+            else:
+                supported = (128,)
+            target = next(s for s in supported if s >= d)
+            pad = target - d
 
             def _pad(t):
                 if t.dtype == torch.float8_e4m3fn:  # F.pad lacks fp8 support
