@@ -1,7 +1,8 @@
-"""Pin the mfma_scale per-lane E8M0 scale-operand semantics on silicon:
-each lane's fragment is one 32-elem k-block, its scale byte goes in bits
-0..7 of the scale operand (byte_sel 0). Oracle applies 2^(s-127) per
-block. Exact match required (fp8 products in fp32 accumulate)."""
+"""Pin the mfma_scale per-lane E8M0 scale-operand semantics on silicon.
+
+Diagnosis mode: set ONE scale entry to 128 (factor 2, all others 127) and
+print which C entries double — that reveals the lane->block mapping of
+the scale operands directly."""
 import sys
 
 sys.path.insert(0, "/root/repo")
@@ -11,23 +12,31 @@ from tree_attention_torch_amd.ops import flash  # noqa: E402
 
 ext = flash._load_extension()
 torch.manual_seed(0)
-a = (torch.randn(32, 64, device="cuda") * 2).to(torch.float8_e4m3fn)
-b = (torch.randn(64, 32, device="cuda") * 2).to(torch.float8_e4m3fn)
-sa = torch.randint(120, 135, (32, 2), device="cuda", dtype=torch.uint8)
-sb = torch.randint(120, 135, (2, 32), device="cuda", dtype=torch.uint8)
-c = ext.probe_mfma_mx_scaled(a, b, sa, sb)
+a = torch.full((32, 64), 1.0, device="cuda").to(torch.float8_e4m3fn)
+b = torch.full((64, 32), 1.0, device="cuda").to(torch.float8_e4m3fn)
+u127 = torch.full((32, 2), 127, device="cuda", dtype=torch.uint8)
+u127b = torch.full((2, 32), 127, device="cuda", dtype=torch.uint8)
 
-af = a.float()
-bf = b.float()
-fa = torch.pow(2.0, sa.float() - 127)        # (32 rows, 2 blocks)
-fb = torch.pow(2.0, sb.float() - 127)        # (2 blocks, 32 cols)
-ref = torch.zeros(32, 32, device="cuda")
-for blk in range(2):
-    asc = af[:, blk * 32:(blk + 1) * 32] * fa[:, blk:blk + 1]
-    bsc = bf[blk * 32:(blk + 1) * 32, :] * fb[blk:blk + 1, :]
-    ref += asc @ bsc
-err = (c - ref).abs().max().item()
-rel = err / ref.abs().max().item()
-print(f"max|err|={err:.3e} rel={rel:.2e}")
-assert rel < 1e-5, "scale semantics mismatch"
-print("MX_SCALED_OK")
+base = ext.probe_mfma_mx_scaled(a, b, u127, u127b)
+print("base C[0,0] (expect 64):", base[0, 0].item())
+
+for (r, blk) in [(0, 0), (0, 1), (5, 0), (5, 1), (17, 0), (31, 1)]:
+    sa = u127.clone()
+    sa[r, blk] = 128
+    c = ext.probe_mfma_mx_scaled(a, b, sa, u127b)
+    ratio = (c / base)
+    rows = sorted(set((ratio > 1.2).nonzero()[:, 0].tolist()))
+    cols = sorted(set((ratio > 1.2).nonzero()[:, 1].tolist()))
+    vals = sorted(set(round(v, 2) for v in ratio[ratio > 1.01].tolist()))
+    print(f"sa[{r}][{blk}]=128 -> boosted rows {rows[:8]} cols "
+          f"{cols[:8]}{'...' if len(cols) > 8 else ''} ratios {vals[:4]}")
+
+for (blk, cidx) in [(0, 0), (1, 0), (0, 9), (1, 30)]:
+    sb = u127b.clone()
+    sb[blk, cidx] = 128
+    c = ext.probe_mfma_mx_scaled(a, b, u127, sb)
+    ratio = (c / base)
+    rows = sorted(set((ratio > 1.2).nonzero()[:, 0].tolist()))
+    cols = sorted(set((ratio > 1.2).nonzero()[:, 1].tolist()))
+    print(f"sb[{blk}][{cidx}]=128 -> boosted rows {rows[:8]}"
+          f"{'...' if len(rows) > 8 else ''} cols {cols[:8]}")
