@@ -40,3 +40,28 @@ for (blk, cidx) in [(0, 0), (1, 0), (0, 9), (1, 30)]:
     cols = sorted(set((ratio > 1.2).nonzero()[:, 1].tolist()))
     print(f"sb[{blk}][{cidx}]=128 -> boosted rows {rows[:8]}"
           f"{'...' if len(rows) > 8 else ''} cols {cols[:8]}")
+
+# full-random verification with a cancellation-aware bound: compare against
+# the magnitude of the LARGEST per-block partial product, not ref.max()
+torch.manual_seed(1)
+a = (torch.randn(32, 64, device="cuda") * 2).to(torch.float8_e4m3fn)
+b = (torch.randn(64, 32, device="cuda") * 2).to(torch.float8_e4m3fn)
+sa = torch.randint(120, 135, (32, 2), device="cuda", dtype=torch.uint8)
+sb = torch.randint(120, 135, (2, 32), device="cuda", dtype=torch.uint8)
+c = ext.probe_mfma_mx_scaled(a, b, sa, sb)
+af, bf = a.float(), b.float()
+fa = torch.pow(2.0, sa.float() - 127)
+fb = torch.pow(2.0, sb.float() - 127)
+ref = torch.zeros(32, 32, device="cuda")
+mag = torch.zeros(32, 32, device="cuda")
+for blk in range(2):
+    asc = af[:, blk * 32:(blk + 1) * 32] * fa[:, blk:blk + 1]
+    bsc = bf[blk * 32:(blk + 1) * 32, :] * fb[blk:blk + 1, :]
+    ref += asc @ bsc
+    mag += asc.abs() @ bsc.abs()
+err = (c - ref).abs()
+rel = (err / mag.clamp(min=1e-30)).max().item()
+print(f"random-scale check: max|err|={err.max().item():.3e} "
+      f"rel-to-magnitude={rel:.2e}")
+assert rel < 1e-5, "scale semantics mismatch"
+print("MX_SCALED_OK")
