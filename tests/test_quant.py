@@ -50,9 +50,10 @@ def test_outlier_channels_recovered():
         rel = ((dq[..., ch] - k[..., ch]).abs() /
                k[..., ch].abs().clamp(min=1e-6)).max()
         assert rel < 0.07, (ch, rel)
-    # every element's abs error bounded by the quantization step of its
-    # row's largest block (block max <= row max)
-    step = k.abs().amax(dim=-1, keepdim=True) * 2.0 ** -8
+    # every element's abs error bounded by the worst e4m3 step of its
+    # row's largest block (rel half-step 2^-4 at the bottom of a binade;
+    # block max <= row max)
+    step = k.abs().amax(dim=-1, keepdim=True) * 2.0 ** -4
     assert ((dq - k).abs() <= step + 1e-6).all()
 
 
