@@ -548,3 +548,75 @@ def test_probe_gen6_softmax_stream(ext):
     c_out, ml_out = ext.probe_gen6_softmax(s_in, ml, 0.125 * 1.44269504)
     assert (c_out[0] == c_out[1]).all(), "c packs differ"
     torch.testing.assert_close(ml_out[0], ml_out[1], rtol=0, atol=0)
+
+
+def test_mx_scale_semantics_pinned(ext):
+    """The mfma_scale per-block E8M0 semantics stay as probed (interleaved
+    k-blocks, per-lane couriers) — tools/check_mx_scales.py as a test."""
+    import importlib.util
+    import os
+
+    spec = importlib.util.spec_from_file_location(
+        "check_mx_scales",
+        os.path.join(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))), "tools", "check_mx_scales.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)  # module-level: imports only
+    worst = max(mod.run("narrow", 1, 125, 130), mod.run("wide", 2, 120, 135))
+    assert worst < 2e-4, worst
+
+
+def test_mx_fp8_attention_outliers(ext):
+    """VERDICT r1 item 4 'done' check: the MX-scaled fp8 attention vs the
+    per-block-scaled oracle, on a KV cache WITH OUTLIERS that breaks the
+    plain unit-scale fp8 path (torch's e4m3 conversion NaNs above 448)."""
+    from tree_attention_torch_amd.ops.flash import local_attention_mx
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+    from tree_attention_torch_amd.quant import (dequantize_k_mx,
+                                                dequantize_v_mx,
+                                                quantize_k_mx, quantize_v_mx)
+
+    torch.manual_seed(7)
+    for (b, hq, hkv, t, tq, causal) in [(1, 4, 4, 512, 256, True),
+                                        (1, 8, 2, 1024, 1024, False),
+                                        (2, 4, 4, 448, 64, True)]:
+        q = torch.randn(b, hq, tq, 128, device="cuda").bfloat16()
+        k = torch.randn(b, hkv, t, 128, device="cuda")
+        v = torch.randn(b, hkv, t, 128, device="cuda")
+        k[..., 13] *= 700.0       # outlier K channel: plain fp8 NaNs
+        v[:, :, 17, :] *= 600.0   # outlier V token
+        assert torch.isnan(k.to(torch.float8_e4m3fn).float()).any()
+        k8, ks = quantize_k_mx(k)
+        v8, vs = quantize_v_mx(v)
+        out, lse = local_attention_mx(q, k8, ks, v8, vs, is_causal=causal,
+                                      q_offset=t - tq)
+        ref_out, ref_lse = flash_res_lse(
+            q.cpu().float(), dequantize_k_mx(k8, ks).cpu(),
+            dequantize_v_mx(v8, vs).cpu(), is_causal=causal, q_offset=t - tq)
+        assert torch.isfinite(out).all()
+        torch.testing.assert_close(lse.cpu(), ref_lse, rtol=2e-2, atol=2e-2)
+        # out tolerance: Q and P are e4m3-quantized in-kernel (their
+        # relative step ~6%); scale errors would be off by POWERS OF TWO,
+        # far outside this band
+        err = (out.cpu() - ref_out).abs()
+        denom = ref_out.abs().clamp(min=1.0)
+        assert (err / denom).max() < 0.08, (err / denom).max()
+
+
+def test_mx_fp8_matches_unit_path_on_tame_data(ext):
+    """On unit-variance data every scale is ~127: the MX path must agree
+    with the unit-scale fp8 prefill closely."""
+    from tree_attention_torch_amd.ops.flash import local_attention_mx
+    from tree_attention_torch_amd.quant import quantize_k_mx, quantize_v_mx
+
+    torch.manual_seed(9)
+    q = torch.randn(1, 4, 512, 128, device="cuda").bfloat16()
+    k = torch.randn(1, 4, 512, 128, device="cuda")
+    v = torch.randn(1, 4, 512, 128, device="cuda")
+    k8, ks = quantize_k_mx(k)
+    v8, vs = quantize_v_mx(v)
+    out_mx, lse_mx = local_attention_mx(q, k8, ks, v8, vs, is_causal=True,
+                                        q_offset=0)
+    out_u, lse_u = ext.flash_attention(q, k8, v8, 128 ** -0.5, True, 0, 0)
+    torch.testing.assert_close(lse_mx, lse_u, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(out_mx, out_u, rtol=6e-2, atol=6e-2)

@@ -181,6 +181,36 @@ def local_attention(
     return reference.flash_res_lse(q, k, v, softmax_scale, is_causal, q_offset, kv_offset)
 
 
+def local_attention_mx(
+    q: torch.Tensor,
+    k8: torch.Tensor,
+    ks: torch.Tensor,
+    v8: torch.Tensor,
+    vs: torch.Tensor,
+    softmax_scale: float | None = None,
+    is_causal: bool = False,
+    q_offset: int = 0,
+    kv_offset: int = 0,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """MX block-scaled fp8 attention: K/V are e4m3 plus per-block E8M0
+    scales from quant.quantize_k_mx / quantize_v_mx. The hardware applies
+    the scales inside mfma_scale_f32_32x32x64_f8f6f4 (block semantics
+    pinned on silicon — tools/check_mx_scales.py), so a KV cache with
+    outlier channels/tokens quantizes at full e4m3 relative precision
+    where plain fp8 saturates to NaN. GPU-only (no CPU oracle dispatch:
+    use quant.dequantize_*_mx + ops.reference for oracles)."""
+    if softmax_scale is None:
+        softmax_scale = 1.0 / math.sqrt(q.shape[-1])
+    ext = _load_extension()
+    if ext is None:
+        raise RuntimeError("MX fp8 attention requires the HIP extension")
+    return ext.flash_attention_fp8_mx(
+        q.contiguous(), k8.contiguous(), ks.contiguous(), v8.contiguous(),
+        vs.contiguous(), float(softmax_scale), bool(is_causal),
+        int(q_offset), int(kv_offset),
+    )
+
+
 # Reference-compatible alias: same name/signature shape as model.py:60.
 def flash_res_lse(
     q: torch.Tensor,
@@ -192,5 +222,5 @@ def flash_res_lse(
     return local_attention(q, k, v, softmax_scale=softmax_scale, is_causal=is_causal)
 
 
-__all__ = ["local_attention", "flash_res_lse", "hip_available",
-           "decode_loop_chunks"]
+__all__ = ["local_attention", "local_attention_mx", "flash_res_lse",
+           "hip_available", "decode_loop_chunks"]

@@ -30,7 +30,8 @@ from __future__ import annotations
 
 import torch
 
-__all__ = ["quantize_k_mx", "dequantize_k_mx", "mx_group_index"]
+__all__ = ["quantize_k_mx", "dequantize_k_mx", "quantize_v_mx",
+           "dequantize_v_mx", "mx_group_index"]
 
 _E4M3_MAX = 448.0
 
@@ -75,3 +76,23 @@ def dequantize_k_mx(k8: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
     gidx = mx_group_index(dims)
     factor = torch.pow(2.0, scales.float() - 127)
     return k8.float() * factor.gather(-1, gidx.expand(k8.shape))
+
+
+def quantize_v_mx(v: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """Quantize V (..., T, D): PV contracts over KEYS, so the scale groups
+    run along the TOKEN axis — per dim, per 64-token window, the two
+    interleaved sets {t: (t>>4)&1 == blk} (same silicon-pinned block
+    structure as K's dims). Returns (v8, scales) with scales uint8 of
+    shape (..., T//32, D) indexed by [token-group][dim], token-group =
+    (t>>6)*2 + ((t>>4)&1). T must be a multiple of 64."""
+    t = v.shape[-2]
+    assert t % 64 == 0, "MX V quantization needs T a multiple of 64"
+    v8t, st = quantize_k_mx(v.transpose(-1, -2).contiguous())
+    return (v8t.transpose(-1, -2).contiguous(),
+            st.transpose(-1, -2).contiguous())
+
+
+def dequantize_v_mx(v8: torch.Tensor, scales: torch.Tensor) -> torch.Tensor:
+    return dequantize_k_mx(
+        v8.transpose(-1, -2).contiguous(),
+        scales.transpose(-1, -2).contiguous()).transpose(-1, -2).contiguous()
