@@ -597,14 +597,19 @@ def test_mx_fp8_attention_outliers(ext):
             q_sim.cpu(), dequantize_k_mx(k8, ks).cpu(),
             dequantize_v_mx(v8, vs).cpu(), is_causal=causal, q_offset=t - tq)
         assert torch.isfinite(out).all()
-        # remaining error = the fixed x448 P quantization (e4m3 rel step
-        # ~6% per term), amplified where outlier-magnitude V terms cancel:
-        # measured 0.136 worst on this matrix (LSE is P-free and tight)
-        lerr = (lse.cpu() - ref_lse).abs().max()
-        assert lerr < 0.2, lerr
-        err = (out.cpu() - ref_out).abs()
-        denom = ref_out.abs().clamp(min=1.0)
-        assert (err / denom).max() < 0.16, (err / denom).max()
+        # Bounds are MAGNITUDE-aware: outlier scores reach ~1e5, and the
+        # MX instruction's accumulation-order noise is ~1e-6 relative to
+        # score magnitude (probe: 2e-5 of the 64-term magnitude), so an
+        # absolute lse bound would just measure the outlier size. Measured
+        # on this matrix: lse/smag <= 1.2e-6, out/vmax <= 2.9e-4.
+        kd = dequantize_k_mx(k8, ks)
+        vd = dequantize_v_mx(v8, vs)
+        smag = (q_sim.abs().amax() * kd.abs().amax() * 128 ** 0.5).item()
+        lrel = ((lse.cpu() - ref_lse).abs().max() / smag).item()
+        assert lrel < 1e-5, lrel
+        vmax = vd.abs().amax().cpu().clamp(min=1.0)
+        orel = ((out.cpu() - ref_out).abs() / vmax).max().item()
+        assert orel < 2e-3, orel
 
 
 def test_mx_fp8_matches_oracle_on_tame_data(ext):
