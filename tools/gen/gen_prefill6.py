@@ -913,7 +913,14 @@ def emit():
     for buf in (0, 1):
         g7 = g7_tile_body(buf * 32768)
         parts.append(f"// gen7 buf{buf} (occ 2, one q-block/wave): {stats(g7)}")
-        parts.append(g7_tile_macro(f"TA_GEN7_TILE_B{buf}", render(g7)))
+        body = render(g7)
+        parts.append(g7_tile_macro(f"TA_GEN7_TILE_B{buf}", body))
+        # fp16 variant: same stream, f16 MFMA + RTZ f16 pack (the only
+        # dtype-dependent instructions)
+        fbody = body.replace("v_mfma_f32_32x32x16_bf16",
+                             "v_mfma_f32_32x32x16_f16") \
+                    .replace("v_cvt_pk_bf16_f32", "v_cvt_pkrtz_f16_f32")
+        parts.append(g7_tile_macro(f"TA_GEN7_TILE_B{buf}_F16", fbody))
     # no-sync ablation: same body, no vmcnt/barriers (garbage output) —
     # isolates the per-tile barrier + DMA-wait share of the wall time
     g8 = g7_tile_body(0)
