@@ -180,3 +180,60 @@ def test_uneven_shards_ws2():
     for p in procs:
         p.join(timeout=120)
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
+def _worker_mx(rank, world_size, port):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from tree_attention_torch_amd.parallel.tree import tree_attention
+        from tree_attention_torch_amd.quant import (dequantize_k_mx,
+                                                    dequantize_v_mx,
+                                                    quantize_k_mx,
+                                                    quantize_v_mx)
+
+        torch.manual_seed(11)
+        b, hkv, d = 1, 2, 128
+        t_total = 256
+        t_local = t_total // world_size
+        q = torch.randn(b, 4, 1, d)
+        k_full = torch.randn(b, hkv, t_total, d)
+        v_full = torch.randn(b, hkv, t_total, d)
+        k_full[..., 9] *= 900.0  # outlier channel: plain fp8 would NaN
+        k = k_full[..., rank * t_local:(rank + 1) * t_local, :]
+        v = v_full[..., rank * t_local:(rank + 1) * t_local, :]
+        k8, ks = quantize_k_mx(k)
+        v8, vs = quantize_v_mx(v)
+        out = tree_attention(q, k8, v8, kv_scales=(ks, vs))
+        # unsharded oracle over the dequantized shards
+        kds, vds = [], []
+        for r in range(world_size):
+            kr = k_full[..., r * t_local:(r + 1) * t_local, :]
+            vr = v_full[..., r * t_local:(r + 1) * t_local, :]
+            k8r, ksr = quantize_k_mx(kr)
+            v8r, vsr = quantize_v_mx(vr)
+            kds.append(dequantize_k_mx(k8r, ksr))
+            vds.append(dequantize_v_mx(v8r, vsr))
+        ref, _ = flash_res_lse(q.float(), torch.cat(kds, dim=-2),
+                               torch.cat(vds, dim=-2))
+        torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-4)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_mx_sharded_ws2():
+    """MX block-scaled fp8 shards through the tree combine (CPU oracle
+    path; the GPU kernel path is tests/test_gpu_kernels.py's mx tests)."""
+    _PORT[0] += 1
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker_mx, args=(r, 2, _PORT[0]))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=120)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]

@@ -61,6 +61,7 @@ def tree_attention(
     return_lse: bool = False,
     kv_offset: int | None = None,
     total_kv: int | None = None,
+    kv_scales: tuple[torch.Tensor, torch.Tensor] | None = None,
 ):
     """Attention over the KV sequence sharded across ranks.
 
@@ -68,6 +69,11 @@ def tree_attention(
         q: (B, Hq, Tq, D) — replicated across ranks.
         k, v: (B, Hkv, T_local, D) — THIS rank's shard; rank r holds global
             positions [r*T_local, (r+1)*T_local).
+        kv_scales: optional (k_scales, v_scales) for an MX block-scaled
+            fp8 shard (quant.quantize_k_mx / quantize_v_mx) — the local
+            partial runs the hardware-scaled kernel on GPU and the
+            dequantized fp32 oracle on CPU; the combine is unchanged
+            (lse algebra is dtype-agnostic).
         is_causal: causal masking in GLOBAL positions; queries are taken to
             be the LAST Tq positions of the global sequence (decode
             semantics: a new token attends to everything before it).
@@ -107,10 +113,25 @@ def tree_attention(
     if q_chunk is None:
         q_chunk = default_q_chunk(tq, q.shape[0], q.shape[1])
 
+    def _local(qc, q_off_c):
+        if kv_scales is None:
+            return local_attention(qc, k, v, softmax_scale, is_causal,
+                                   q_off_c, kv_offset)
+        ks, vs = kv_scales
+        if qc.device.type == "cuda":
+            from ..ops.flash import local_attention_mx
+
+            return local_attention_mx(qc, k, ks, v, vs, softmax_scale,
+                                      is_causal, q_off_c, kv_offset)
+        from ..ops.reference import flash_res_lse as oracle
+        from ..quant import dequantize_k_mx, dequantize_v_mx
+
+        return oracle(qc.float(), dequantize_k_mx(k, ks),
+                      dequantize_v_mx(v, vs), softmax_scale, is_causal,
+                      q_off_c, kv_offset)
+
     if tq <= q_chunk:
-        out_l, lse_l = local_attention(
-            q, k, v, softmax_scale, is_causal, q_offset, kv_offset
-        )
+        out_l, lse_l = _local(q, q_offset)
         out, lse = tree_combine(out_l, lse_l, strategy=combine, group=group)
         return (out, lse) if return_lse else out
 
@@ -122,9 +143,7 @@ def tree_attention(
     for c in range(n_chunks):
         lo = c * q_chunk
         hi = min(tq, lo + q_chunk)
-        out_c, lse_c = local_attention(
-            q[..., lo:hi, :], k, v, softmax_scale, is_causal, q_offset + lo, kv_offset
-        )
+        out_c, lse_c = _local(q[..., lo:hi, :], q_offset + lo)
         outs.append(None)
         lses.append(None)
         if overlap:
