@@ -612,6 +612,54 @@ def test_mx_fp8_attention_outliers(ext):
         assert orel < 2e-3, orel
 
 
+def test_mx_decode_vs_oracle(ext):
+    """MX decode route (G*Tq <= 16): split-KV kernel with a dequantizing
+    stage — fp8 HBM bandwidth, bf16 compute (the full-precision softmax/PV
+    of the decode kernel; no in-kernel Q or P quantization). The oracle is
+    the fp32 flash reference over the bf16-ROUNDED dequantized cache (the
+    stage writes bf16 LDS images), so the comparison isolates the stage's
+    scale application; includes an outlier cache that NaNs plain fp8."""
+    from tree_attention_torch_amd.ops.flash import local_attention_mx
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+    from tree_attention_torch_amd.quant import (dequantize_k_mx,
+                                                dequantize_v_mx,
+                                                quantize_k_mx, quantize_v_mx)
+
+    torch.manual_seed(11)
+    for (b, hq, hkv, t, tq, outlier) in [
+            (1, 8, 8, 4096, 1, False),     # MHA decode
+            (1, 8, 1, 16384, 2, False),    # GQA 8, tq=2 (G*Tq = 16), splits
+            (2, 4, 1, 4032, 1, True),      # outliers; Tkv % 64 but not 128
+    ]:
+        q = torch.randn(b, hq, tq, 128, device="cuda").bfloat16()
+        k = torch.randn(b, hkv, t, 128, device="cuda")
+        v = torch.randn(b, hkv, t, 128, device="cuda")
+        if outlier:
+            k[..., 13] *= 700.0
+            v[:, :, 17, :] *= 600.0
+            assert torch.isnan(k.to(torch.float8_e4m3fn).float()).any()
+        k8, ks = quantize_k_mx(k)
+        v8, vs = quantize_v_mx(v)
+        out, lse = local_attention_mx(q, k8, ks, v8, vs, is_causal=True,
+                                      q_offset=t - tq)
+        assert torch.isfinite(out).all()
+        kd = dequantize_k_mx(k8, ks).bfloat16().float().cpu()
+        vd = dequantize_v_mx(v8, vs).bfloat16().float().cpu()
+        ref_out, ref_lse = flash_res_lse(q.float().cpu(), kd, vd,
+                                         is_causal=True, q_offset=t - tq)
+        if outlier:
+            smag = (q.float().abs().amax().cpu() * kd.abs().amax()
+                    * 128 ** 0.5).item()
+            assert ((lse.cpu() - ref_lse).abs().max() / smag).item() < 1e-5
+            vmax = vd.abs().amax().clamp(min=1.0)
+            assert ((out.cpu() - ref_out).abs() / vmax).max().item() < 2e-3
+        else:
+            torch.testing.assert_close(lse.cpu(), ref_lse, rtol=1e-2,
+                                       atol=1e-2)
+            torch.testing.assert_close(out.cpu(), ref_out, rtol=2e-2,
+                                       atol=2e-2)
+
+
 def test_mx_fp8_matches_oracle_on_tame_data(ext):
     """On unit-variance data the quantizer legitimately UP-scales each
     block into e4m3's range (scales ~121), so the right check is against
