@@ -204,10 +204,20 @@ def local_attention_mx(
     ext = _load_extension()
     if ext is None:
         raise RuntimeError("MX fp8 attention requires the HIP extension")
+    decode_shaped = (q.shape[1] // k8.shape[1]) * q.shape[2] <= 16
+
+    def _rows_contig(t):  # per-head rows packed; head stride free
+        return (t.stride(3) == 1 and t.stride(2) == t.shape[3]
+                and t.stride(0) == t.stride(1) * t.shape[1])
+
+    if not (decode_shaped and all(map(_rows_contig, (k8, ks, v8, vs)))):
+        # the prefill kernel wants packed tensors; the decode kernel takes
+        # per-head byte strides, so session-cache VIEWS pass zero-copy
+        k8, ks = k8.contiguous(), ks.contiguous()
+        v8, vs = v8.contiguous(), vs.contiguous()
     return ext.flash_attention_fp8_mx(
-        q.contiguous(), k8.contiguous(), ks.contiguous(), v8.contiguous(),
-        vs.contiguous(), float(softmax_scale), bool(is_causal),
-        int(q_offset), int(kv_offset),
+        q.contiguous(), k8, ks, v8, vs, float(softmax_scale),
+        bool(is_causal), int(q_offset), int(kv_offset),
     )
 
 

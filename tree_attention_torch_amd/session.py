@@ -27,7 +27,7 @@ import torch
 import torch.distributed as dist
 
 from .ops.flash import local_attention
-from .parallel.combine import tree_combine
+from .parallel.combine import combine_partials, tree_combine
 
 __all__ = ["DecodeSession"]
 
@@ -61,12 +61,44 @@ class DecodeSession:
             self.rank, self.world = 0, 1
         self.group = group
         self.block = block
-        td = _DTYPES[kv_dtype] if isinstance(kv_dtype, str) else kv_dtype
+        self.mx = isinstance(kv_dtype, str) and kv_dtype in ("mx", "fp8_mx")
         blocks_total = (max_tokens + block - 1) // block + 1
         local_cap = ((blocks_total + self.world - 1) // self.world) * block
-        self.k = torch.empty(batch, kv_heads, local_cap, head_dim,
-                             dtype=td, device=self.device)
-        self.v = torch.empty_like(self.k)
+        if self.mx:
+            # MX block-scaled fp8 cache (quant.py layout): e4m3 payload +
+            # per-block E8M0 scales, quantized in closed 64-token WINDOWS
+            # (the scale-block granularity). Tokens land in a bf16 staging
+            # tail first; each full window quantizes into the main cache.
+            # attend() merges the quantized-prefix partial (hardware
+            # mfma_scale decode kernel, ZERO-COPY strided views of this
+            # preallocated cache) with the tail partial (bf16 kernel) via
+            # the standard lse algebra — so per-token accuracy of the
+            # newest <=63 tokens is full bf16 and the long prefix streams
+            # at fp8 bandwidth with outlier-robust block scales.
+            if head_dim != 128:
+                raise ValueError("MX KV cache requires head_dim=128")
+            if block % 64:
+                raise ValueError("MX KV cache requires block % 64 == 0")
+            self.k = torch.empty(batch, kv_heads, local_cap, head_dim,
+                                 dtype=torch.float8_e4m3fn,
+                                 device=self.device)
+            self.v = torch.empty_like(self.k)
+            self.ks = torch.empty(batch, kv_heads, local_cap, 4,
+                                  dtype=torch.uint8, device=self.device)
+            self.vs = torch.empty(batch, kv_heads, local_cap // 32,
+                                  head_dim, dtype=torch.uint8,
+                                  device=self.device)
+            self.k_tail = torch.empty(batch, kv_heads, 64, head_dim,
+                                      dtype=torch.bfloat16,
+                                      device=self.device)
+            self.v_tail = torch.empty_like(self.k_tail)
+            self.q_len = 0     # quantized prefix length (multiple of 64)
+            self.tail_len = 0  # bf16 staging-tail tokens (< 64)
+        else:
+            td = _DTYPES[kv_dtype] if isinstance(kv_dtype, str) else kv_dtype
+            self.k = torch.empty(batch, kv_heads, local_cap, head_dim,
+                                 dtype=td, device=self.device)
+            self.v = torch.empty_like(self.k)
         self.total = 0       # global tokens appended so far
         self.local_len = 0   # tokens in THIS rank's shard
 
@@ -85,14 +117,108 @@ class DecodeSession:
                     f"DecodeSession capacity exceeded: local shard full at "
                     f"{self.local_len} tokens (grow max_tokens)"
                 )
-            self.k[:, :, self.local_len : self.local_len + 1] = \
-                k_new.to(self.k.dtype)
-            self.v[:, :, self.local_len : self.local_len + 1] = \
-                v_new.to(self.v.dtype)
+            if self.mx:
+                self._store_mx(k_new, v_new)
+            else:
+                self.k[:, :, self.local_len : self.local_len + 1] = \
+                    k_new.to(self.k.dtype)
+                self.v[:, :, self.local_len : self.local_len + 1] = \
+                    v_new.to(self.v.dtype)
             self.local_len += 1
         self.total += 1
 
+    def _store_mx(self, kc: torch.Tensor, vc: torch.Tensor) -> None:
+        """Store owned tokens (B, Hkv, n, D) into the MX cache: fill the
+        bf16 tail, quantize each full 64-token window (bulk windows
+        quantize directly when the tail is empty)."""
+        from .quant import quantize_k_mx, quantize_v_mx
+
+        n, i = kc.shape[2], 0
+        while i < n:
+            if self.tail_len > 0 or n - i < 64:
+                m = min(64 - self.tail_len, n - i)
+                self.k_tail[:, :, self.tail_len : self.tail_len + m] = \
+                    kc[:, :, i : i + m].to(torch.bfloat16)
+                self.v_tail[:, :, self.tail_len : self.tail_len + m] = \
+                    vc[:, :, i : i + m].to(torch.bfloat16)
+                self.tail_len += m
+                i += m
+                if self.tail_len == 64:
+                    k8w, ksw = quantize_k_mx(self.k_tail.float())
+                    v8w, vsw = quantize_v_mx(self.v_tail.float())
+                    q0 = self.q_len
+                    self.k[:, :, q0 : q0 + 64] = k8w
+                    self.ks[:, :, q0 : q0 + 64] = ksw
+                    self.v[:, :, q0 : q0 + 64] = v8w
+                    self.vs[:, :, q0 // 32 : q0 // 32 + 2] = vsw
+                    self.q_len += 64
+                    self.tail_len = 0
+            else:
+                m = (n - i) // 64 * 64
+                # round through bf16 first so bulk-prefilled windows
+                # quantize the SAME stream as tail-flushed ones (appends
+                # stage in the bf16 tail before quantization)
+                k8w, ksw = quantize_k_mx(
+                    kc[:, :, i : i + m].to(torch.bfloat16).float())
+                v8w, vsw = quantize_v_mx(
+                    vc[:, :, i : i + m].to(torch.bfloat16).float())
+                q0 = self.q_len
+                self.k[:, :, q0 : q0 + m] = k8w
+                self.ks[:, :, q0 : q0 + m] = ksw
+                self.v[:, :, q0 : q0 + m] = v8w
+                self.vs[:, :, q0 // 32 : (q0 + m) // 32] = vsw
+                self.q_len += m
+                i += m
+
+    def _local_partial_mx(self, q: torch.Tensor,
+                          softmax_scale: float | None):
+        """MX-cache local partial: quantized-prefix partial (hardware MX
+        decode kernel on GPU / dequantized oracle on CPU) merged with the
+        bf16 staging-tail partial through the stable lse algebra."""
+        parts = []
+        if self.q_len > 0:
+            if q.device.type == "cuda":
+                from .ops.flash import local_attention_mx
+
+                parts.append(local_attention_mx(
+                    q, self.k[:, :, : self.q_len],
+                    self.ks[:, :, : self.q_len],
+                    self.v[:, :, : self.q_len],
+                    self.vs[:, :, : self.q_len // 32], softmax_scale))
+            else:
+                from .quant import dequantize_k_mx, dequantize_v_mx
+
+                kd = dequantize_k_mx(self.k[:, :, : self.q_len].contiguous(),
+                                     self.ks[:, :, : self.q_len].contiguous())
+                vd = dequantize_v_mx(
+                    self.v[:, :, : self.q_len].contiguous(),
+                    self.vs[:, :, : self.q_len // 32].contiguous())
+                parts.append(local_attention(q.float(), kd, vd,
+                                             softmax_scale))
+        if self.tail_len > 0:
+            kt = self.k_tail[:, :, : self.tail_len].contiguous()
+            vt = self.v_tail[:, :, : self.tail_len].contiguous()
+            if q.device.type == "cuda":
+                parts.append(local_attention(q, kt.to(q.dtype),
+                                             vt.to(q.dtype), softmax_scale))
+            else:
+                parts.append(local_attention(q.float(), kt.float(),
+                                             vt.float(), softmax_scale))
+        if not parts:
+            b, hq, tq, d = q.shape
+            out_l = torch.zeros(b, hq, tq, d, dtype=torch.float32,
+                                device=q.device)
+            lse_l = torch.full((b, hq, tq), float("-inf"),
+                               dtype=torch.float32, device=q.device)
+            return out_l, lse_l
+        if len(parts) == 1:
+            return parts[0]
+        return combine_partials(torch.stack([p[0] for p in parts]),
+                                torch.stack([p[1] for p in parts]))
+
     def _local_partial(self, q: torch.Tensor, softmax_scale: float | None):
+        if self.mx:
+            return self._local_partial_mx(q, softmax_scale)
         # zero-copy eligibility mirrors the binding's TORCH_CHECKs
         # (bindings.cpp flash_attention_cache): D=128; q bf16/fp16; cache
         # dtype == q dtype or fp8 cache under a bf16 q; (Hq/Hkv)*Tq <= 16
@@ -172,6 +298,12 @@ class DecodeSession:
         replay()'s return value is the fresh combined output.
         """
         assert self.device.type == "cuda"
+        if self.mx:
+            raise NotImplementedError(
+                "graphed_attend: the MX cache path is two kernels + a "
+                "python merge with per-call allocations; use eager "
+                "attend() (the MX decode kernel itself is ~0.2 ms at "
+                "128K — not launch-bound)")
         # warmup on a side stream (allocator + kernels), then capture
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
@@ -220,10 +352,14 @@ class DecodeSession:
                         f"DecodeSession capacity exceeded during prefill at "
                         f"{self.local_len}+{n} tokens (grow max_tokens)"
                     )
-                self.k[:, :, self.local_len : self.local_len + n] = \
-                    k_seq[:, :, t : t + n].to(self.k.dtype)
-                self.v[:, :, self.local_len : self.local_len + n] = \
-                    v_seq[:, :, t : t + n].to(self.v.dtype)
+                if self.mx:
+                    self._store_mx(k_seq[:, :, t : t + n],
+                                   v_seq[:, :, t : t + n])
+                else:
+                    self.k[:, :, self.local_len : self.local_len + n] = \
+                        k_seq[:, :, t : t + n].to(self.k.dtype)
+                    self.v[:, :, self.local_len : self.local_len + n] = \
+                        v_seq[:, :, t : t + n].to(self.v.dtype)
                 self.local_len += n
             self.total += n
             t += n
