@@ -23,6 +23,15 @@ sess.prefill(ks, vs)
 qg = torch.randn(1, 8, 1, 128, device=dev).bfloat16()
 replay, out_g = sess.graphed_attend(qg)
 
+# MX serving session (hardware-scale decode over the quantized cache,
+# bf16 staging tail, outlier channel alive the whole soak)
+mx_sess = DecodeSession(1, 4, 128, max_tokens=1 << 16, device=dev,
+                        kv_dtype="mx", block=256)
+kmx = torch.randn(1, 4, 8192, 128, device=dev)
+kmx[..., 13] *= 700.0
+mx_sess.prefill(kmx, kmx)
+qmx = torch.randn(1, 16, 1, 128, device=dev).bfloat16()
+
 t0 = time.time()
 it = 0
 checks = 0
@@ -38,6 +47,12 @@ while time.time() - t0 < secs:
     sess.sync_len()
     replay()
     assert torch.isfinite(out_g).all()
+    if it % 5 == 0:
+        kn4 = torch.randn(1, 4, 1, 128, device=dev)
+        mx_sess.append(kn4 * (1 + 700.0 * (torch.rand_like(kn4) < 0.01)),
+                       kn4)
+        omx = mx_sess.attend(qmx)
+        assert torch.isfinite(omx).all()
     if it % 50 == 0:
         # numerics spot-check on a fresh small case
         torch.manual_seed(1000 + it)
