@@ -159,6 +159,45 @@ def test_rccl_decode_session():
     _spawn(_session_worker, (), NGPU)
 
 
+def _mx_session_worker(rank, world, port):
+    _env(rank, world, port)
+    import torch as t
+    from tree_attention_torch_amd.ops.flash import local_attention
+    from tree_attention_torch_amd.parallel.pg import cleanup, setup
+    from tree_attention_torch_amd.session import DecodeSession
+
+    device = t.device(f"cuda:{rank}")
+    setup(rank, world)
+    try:
+        g = t.Generator(device=device).manual_seed(13)
+        sess = DecodeSession(1, 8, 128, max_tokens=4096, device=device,
+                             kv_dtype="mx", block=64)
+        kseq = t.randn(1, 8, 1500, 128, generator=g, device=device)
+        vseq = t.randn(1, 8, 1500, 128, generator=g, device=device)
+        kseq[..., 13] *= 700.0  # outlier channel (plain fp8 would NaN)
+        q = t.randn(1, 8, 1, 128, generator=g, device=device,
+                    dtype=t.bfloat16)
+        sess.prefill(kseq, vseq)
+        for i in range(3):  # exercise tail appends across ranks
+            sess.append(kseq[:, :, i : i + 1], vseq[:, :, i : i + 1])
+        out = sess.attend(q)
+        t.cuda.synchronize()
+        assert t.isfinite(out).all()
+        if rank == 0:
+            ku = t.cat([kseq, kseq[:, :, :3]], dim=2).bfloat16()
+            vu = t.cat([vseq, vseq[:, :, :3]], dim=2).bfloat16()
+            ref, _ = local_attention(q, ku, vu)
+            # MX quantization + in-kernel Q/P quantization: fp8-class bar
+            t.testing.assert_close(out, ref, rtol=5e-2, atol=5e-2)
+    finally:
+        cleanup()
+
+
+@multi
+def test_rccl_mx_session():
+    _spawn(_mx_session_worker, (), NGPU)
+
+
 def _graphed_session_worker(rank, world, port):
     _env(rank, world, port)
     import torch as t
