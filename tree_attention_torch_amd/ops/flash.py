@@ -204,6 +204,10 @@ def local_attention_mx(
     ext = _load_extension()
     if ext is None:
         raise RuntimeError("MX fp8 attention requires the HIP extension")
+    if q.dtype == torch.float16:
+        # the MX kernels quantize Q to e4m3 in-prologue; a bf16 round-trip
+        # of an fp16 query is lossless relative to that quantization
+        q = q.to(torch.bfloat16)
     decode_shaped = (q.shape[1] // k8.shape[1]) * q.shape[2] <= 16
 
     def _rows_contig(t):  # per-head rows packed; head stride free
