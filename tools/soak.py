@@ -25,8 +25,9 @@ replay, out_g = sess.graphed_attend(qg)
 
 # MX serving session (hardware-scale decode over the quantized cache,
 # bf16 staging tail, outlier channel alive the whole soak)
-mx_sess = DecodeSession(1, 4, 128, max_tokens=1 << 16, device=dev,
-                        kv_dtype="mx", block=256)
+mx_sess = DecodeSession(1, 4, 128, max_tokens=1 << 21, device=dev,
+                        kv_dtype="mx", block=256)  # 2M-token headroom:
+# a 600 s soak appends ~70k tokens (1<<16 overflowed at ~550 s)
 kmx = torch.randn(1, 4, 8192, 128, device=dev)
 kmx[..., 13] *= 700.0
 mx_sess.prefill(kmx, kmx)
