@@ -102,3 +102,30 @@ class TestCombineStrategy:
     def test_boundary(self):
         assert auto_strategy((1 << 18) - 1) == "allgather"
         assert auto_strategy(1 << 18) == "allreduce"
+
+
+def test_mx_decode_shaped_pin():
+    """MX route decision: G*Tq <= 16 -> hardware-scale decode kernel."""
+    from tree_attention_torch_amd.ops.flash import mx_decode_shaped
+
+    assert mx_decode_shaped(32, 32, 1)        # MHA decode
+    assert mx_decode_shaped(8, 1, 2)          # GQA8, tq=2 (exactly 16)
+    assert not mx_decode_shaped(8, 1, 3)      # 24 rows -> prefill
+    assert not mx_decode_shaped(32, 32, 17)   # spec batch -> prefill
+    assert mx_decode_shaped(16, 16, 16)       # 16 rows boundary
+
+
+def test_mx_rows_contig_pin():
+    """Zero-copy eligibility: session-cache views qualify, transposes and
+    gathered tensors do not."""
+    import torch
+
+    from tree_attention_torch_amd.ops.flash import _rows_contig
+
+    cache = torch.zeros(2, 4, 256, 128, dtype=torch.uint8)
+    assert _rows_contig(cache)
+    assert _rows_contig(cache[:, :, :192])          # sliced view (live len)
+    assert not _rows_contig(cache.transpose(2, 3))  # rows not packed
+    assert not _rows_contig(cache[:, :, :, :64])   # row prefix: holes
+    scales = torch.zeros(2, 4, 256, 4, dtype=torch.uint8)
+    assert _rows_contig(scales[:, :, :64])

@@ -181,6 +181,22 @@ def local_attention(
     return reference.flash_res_lse(q, k, v, softmax_scale, is_causal, q_offset, kv_offset)
 
 
+def mx_decode_shaped(hq: int, hkv: int, tq: int) -> bool:
+    """MX route decision (pure; pinned by tests/test_routing.py): group
+    rows G*Tq <= 16 fit one MFMA-M batch, so the split-KV hardware-scale
+    DECODE kernel runs (zero-copy-capable, per-head byte strides);
+    anything larger goes to the MX prefill kernel (packed tensors)."""
+    return (hq // hkv) * tq <= 16
+
+
+def _rows_contig(t: "torch.Tensor") -> bool:
+    """Per-head rows packed (stride(3)==1, stride(2)==row length) with a
+    free head stride — the layout the MX decode kernel binds ZERO-COPY
+    (sliced views of a preallocated DecodeSession cache)."""
+    return (t.stride(3) == 1 and t.stride(2) == t.shape[3]
+            and t.stride(0) == t.stride(1) * t.shape[1])
+
+
 def local_attention_mx(
     q: torch.Tensor,
     k8: torch.Tensor,
@@ -208,12 +224,7 @@ def local_attention_mx(
         # the MX kernels quantize Q to e4m3 in-prologue; a bf16 round-trip
         # of an fp16 query is lossless relative to that quantization
         q = q.to(torch.bfloat16)
-    decode_shaped = (q.shape[1] // k8.shape[1]) * q.shape[2] <= 16
-
-    def _rows_contig(t):  # per-head rows packed; head stride free
-        return (t.stride(3) == 1 and t.stride(2) == t.shape[3]
-                and t.stride(0) == t.stride(1) * t.shape[1])
-
+    decode_shaped = mx_decode_shaped(q.shape[1], k8.shape[1], q.shape[2])
     if not (decode_shaped and all(map(_rows_contig, (k8, ks, v8, vs)))):
         # the prefill kernel wants packed tensors; the decode kernel takes
         # per-head byte strides, so session-cache VIEWS pass zero-copy
