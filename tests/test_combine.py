@@ -101,3 +101,48 @@ def test_single_shard_identity():
     out, lse = combine_partials(o.unsqueeze(0), l.unsqueeze(0))
     torch.testing.assert_close(out, o)
     torch.testing.assert_close(lse, l)
+
+
+def test_combine_partials_property_random_splits():
+    """Hypothesis property: combining ANY partition of the KV into S
+    partials reproduces the joint attention (the algebra is exact up to
+    fp32 roundoff for every split arrangement, not just equal shards)."""
+    from hypothesis import given, settings, strategies as st
+
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+    from tree_attention_torch_amd.parallel.combine import combine_partials
+
+    @settings(max_examples=20, deadline=None)
+    @given(st.integers(0, 2**31 - 1),
+           st.lists(st.integers(1, 64), min_size=1, max_size=6))
+    def run(seed, sizes):
+        torch.manual_seed(seed)
+        t = sum(sizes)
+        q = torch.randn(1, 2, 1, 32)
+        k = torch.randn(1, 2, t, 32)
+        v = torch.randn(1, 2, t, 32)
+        ref, ref_lse = flash_res_lse(q, k, v)
+        outs, lses = [], []
+        lo = 0
+        for n in sizes:
+            o, l = flash_res_lse(q, k[:, :, lo : lo + n],
+                                 v[:, :, lo : lo + n])
+            outs.append(o)
+            lses.append(l)
+            lo += n
+        out, lse = combine_partials(torch.stack(outs), torch.stack(lses))
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+        torch.testing.assert_close(lse, ref_lse, rtol=1e-5, atol=1e-5)
+
+    run()
+
+
+def test_combine_partials_all_empty():
+    """All partials empty (-inf lse): out = 0, lse = -inf, no NaN."""
+    from tree_attention_torch_amd.parallel.combine import combine_partials
+
+    outs = torch.zeros(3, 1, 2, 1, 16)
+    lses = torch.full((3, 1, 2, 1), float("-inf"))
+    out, lse = combine_partials(outs, lses)
+    assert torch.isfinite(out).all() and (out == 0).all()
+    assert torch.isinf(lse).all() and (lse < 0).all()

@@ -75,3 +75,50 @@ def test_roundtrip_exact_powers():
     k2 = k * 2.0 ** 20   # whole tensor far above plain-fp8 range
     k8b, sb = quantize_k_mx(k2)
     torch.testing.assert_close(dequantize_k_mx(k8b, sb), k2)
+
+
+def test_quantization_error_bound_property():
+    """Hypothesis property: for any finite cache, the MX round-trip error
+    is bounded by e4m3's relative precision (2^-3 of the block max after
+    E8M0 power-of-two scaling can at most double the spacing) — every
+    element within 1/8 of its BLOCK's absmax, zeros preserved."""
+    from hypothesis import given, settings, strategies as st
+
+    from tree_attention_torch_amd.quant import dequantize_k_mx, quantize_k_mx
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.integers(0, 2**31 - 1), st.floats(-30.0, 30.0))
+    def run(seed, log2scale):
+        torch.manual_seed(seed)
+        k = torch.randn(1, 1, 64, 128, dtype=torch.float32)
+        k *= 2.0 ** log2scale
+        k8, ks = quantize_k_mx(k)
+        kd = dequantize_k_mx(k8, ks)
+        assert torch.isfinite(kd).all()
+        # per-block bound: group = (d>>6)*2 + ((d>>4)&1) over dims
+        d = torch.arange(128)
+        g = ((d >> 6) * 2 + ((d >> 4) & 1))
+        for blk in range(4):
+            cols = g == blk
+            sub, subd = k[..., cols], kd[..., cols]
+            bound = sub.abs().amax(dim=-1, keepdim=True) / 8.0 + 1e-30
+            assert ((sub - subd).abs() <= bound + 1e-6 * sub.abs()).all()
+        assert (kd[k == 0] == 0).all()
+
+    run()
+
+
+def test_v_quantizer_transpose_consistency():
+    """quantize_v_mx is quantize_k_mx on the transpose: dequantized V must
+    equal the transposed dequantized-K of the transposed input."""
+    from tree_attention_torch_amd.quant import (dequantize_k_mx,
+                                                dequantize_v_mx,
+                                                quantize_k_mx, quantize_v_mx)
+
+    torch.manual_seed(4)
+    v = torch.randn(2, 3, 128, 128) * 40.0
+    v8, vs = quantize_v_mx(v)
+    vd = dequantize_v_mx(v8, vs)
+    k8t, kst = quantize_k_mx(v.transpose(-1, -2).contiguous())
+    kdt = dequantize_k_mx(k8t, kst).transpose(-1, -2)
+    torch.testing.assert_close(vd, kdt)
