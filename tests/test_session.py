@@ -292,3 +292,41 @@ def test_session_mx_gpu():
     ref, _ = flash_res_lse(q.float().cpu(), ku, vu)
     vmax = vu.abs().amax().clamp(min=1.0)
     assert ((out.cpu() - ref).abs() / vmax).max().item() < 4e-2
+
+
+def test_session_block_cyclic_ownership_property():
+    """Hypothesis property (pure accounting): for any world size, block
+    size and append sequence, each rank's local_len equals the number of
+    global tokens whose block-cyclic owner is that rank, and the mx
+    quantized-prefix/tail split always satisfies q_len % 64 == 0,
+    tail_len < 64, q_len + tail_len == local_len."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=30, deadline=None)
+    @given(st.integers(1, 8), st.sampled_from([64, 128, 256]),
+           st.lists(st.integers(1, 200), min_size=1, max_size=5))
+    def run(world, block, chunks):
+        total = sum(chunks)
+        sessions = [
+            DecodeSession(1, 1, 128, max_tokens=max(total + block, 1024),
+                          device="cpu", kv_dtype="mx", block=block)
+            for _ in range(world)
+        ]
+        # monkey-set rank/world (no process group on CPU single-proc)
+        for r, s in enumerate(sessions):
+            s.rank, s.world = r, world
+        ks = torch.randn(1, 1, total, 128)
+        t = 0
+        for n in chunks:
+            for s in sessions:
+                s.prefill(ks[:, :, t : t + n], ks[:, :, t : t + n])
+            t += n
+        for r, s in enumerate(sessions):
+            expect = sum(1 for g in range(total)
+                         if (g // block) % world == r)
+            assert s.local_len == expect, (r, s.local_len, expect)
+            assert s.q_len % 64 == 0 and 0 <= s.tail_len < 64
+            assert s.q_len + s.tail_len == s.local_len
+            assert s.total == total
+
+    run()

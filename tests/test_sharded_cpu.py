@@ -237,3 +237,34 @@ def test_mx_sharded_ws2():
     for p in procs:
         p.join(timeout=120)
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
+def test_mx_kv_offset_causal_cpu():
+    """tree_attention(kv_scales=...) with explicit kv_offset/total_kv
+    (uneven-shard plumbing) on the CPU oracle path: causal positions are
+    GLOBAL, so a middle shard must mask exactly like the dequantized
+    oracle with the same offsets."""
+    import torch
+
+    from tree_attention_torch_amd.ops.reference import flash_res_lse
+    from tree_attention_torch_amd.parallel.tree import tree_attention
+    from tree_attention_torch_amd.quant import (dequantize_k_mx,
+                                                dequantize_v_mx,
+                                                quantize_k_mx, quantize_v_mx)
+
+    torch.manual_seed(21)
+    b, h, d = 1, 2, 128
+    t_local, kv_off, total = 128, 256, 512
+    q = torch.randn(b, h, 64, d).bfloat16()   # last 64 global positions
+    k = torch.randn(b, h, t_local, d)
+    v = torch.randn(b, h, t_local, d)
+    k8, ks = quantize_k_mx(k)
+    v8, vs = quantize_v_mx(v)
+    out, lse = tree_attention(q, k8, v8, is_causal=True,
+                              kv_offset=kv_off, total_kv=total,
+                              kv_scales=(ks, vs), return_lse=True)
+    ref, ref_lse = flash_res_lse(q.float(), dequantize_k_mx(k8, ks),
+                                 dequantize_v_mx(v8, vs), is_causal=True,
+                                 q_offset=total - 64, kv_offset=kv_off)
+    torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(lse, ref_lse, rtol=1e-5, atol=1e-5)
