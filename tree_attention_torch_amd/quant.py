@@ -1,5 +1,11 @@
 """MX (block-scaled) fp8 quantization for the KV cache.
 
+Reference parity note: the reference (/root/reference/model.py) has no
+quantization of any kind — this module is a capability EXTENSION of the
+reference's fp16 attention (model.py:51-53) for the MI355X serving
+target (BASELINE.json config 5's fp8 KV path, hardened against outlier
+channels/tokens that plain e4m3 saturates or NaNs on).
+
 The gfx950 `mfma_scale_f32_32x32x64_f8f6f4` instruction dequantizes its
 fp8 operands with per-32-element E8M0 scales in hardware. The scale-block
 semantics were pinned on silicon (tools/check_mx_scales.py): within each
